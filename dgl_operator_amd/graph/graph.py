@@ -1,0 +1,218 @@
+"""Graph containers: CSC/CSR homogeneous graph and bipartite message-flow Block.
+
+Design (MI355X-first):
+  * The forward message-passing structure is CSC (per-destination in-edges):
+    ``update_all`` aggregates messages over in-edges, so the SpMM kernel walks
+    csc_indptr/csc_indices. The transposed structure (CSR, per-source
+    out-edges) is built lazily for the backward pass.
+  * Edge-order mapping ``csc_eids``/``csr_eids`` translates a position in the
+    CSC/CSR arrays back to the original COO edge id so per-edge data
+    (weights, attention scores) line up.
+  * Blocks keep DGL's convention that the first ``num_dst`` source nodes ARE
+    the destination nodes (needed by SAGEConv's self-feature path).
+
+Reference behavior being matched (not ported): DGLGraph/``dgl.to_block`` as
+used by /root/reference/examples/GraphSAGE/code/3_message_passing.py and
+examples/GraphSAGE_dist/code/train_dist.py:52-70.
+"""
+from __future__ import annotations
+
+from typing import List, Optional, Sequence, Tuple
+
+import torch
+
+
+def _coo_to_compressed(
+    rows: torch.Tensor, cols: torch.Tensor, num_rows: int
+) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    """Sort edges by ``rows`` and build indptr. Returns (indptr, sorted_cols, eids)."""
+    perm = torch.argsort(rows, stable=True)
+    sorted_rows = rows[perm]
+    indptr = torch.zeros(num_rows + 1, dtype=torch.int64, device=rows.device)
+    ones = torch.ones_like(sorted_rows)
+    indptr.scatter_add_(0, sorted_rows + 1, ones)
+    indptr = torch.cumsum(indptr, dim=0)
+    return indptr, cols[perm], perm
+
+
+class Graph:
+    """Homogeneous directed graph. Edges point src -> dst; messages flow along edges."""
+
+    def __init__(
+        self,
+        src: torch.Tensor,
+        dst: torch.Tensor,
+        num_nodes: Optional[int] = None,
+    ):
+        if src.dtype != torch.int64:
+            src = src.to(torch.int64)
+        if dst.dtype != torch.int64:
+            dst = dst.to(torch.int64)
+        self._src = src
+        self._dst = dst
+        if num_nodes is None:
+            num_nodes = int(torch.max(torch.stack([src.max(), dst.max()])).item()) + 1 if src.numel() else 0
+        self._num_nodes = int(num_nodes)
+        # CSC: per-dst in-neighbors (forward aggregation structure)
+        self._csc: Optional[Tuple[torch.Tensor, torch.Tensor, torch.Tensor]] = None
+        # CSR: per-src out-neighbors (backward / transposed structure)
+        self._csr: Optional[Tuple[torch.Tensor, torch.Tensor, torch.Tensor]] = None
+        self.ndata = {}
+        self.edata = {}
+
+    # -- structure ---------------------------------------------------------
+    @property
+    def num_nodes(self) -> int:
+        return self._num_nodes
+
+    @property
+    def num_edges(self) -> int:
+        return self._src.numel()
+
+    @property
+    def device(self) -> torch.device:
+        return self._src.device
+
+    def edges(self) -> Tuple[torch.Tensor, torch.Tensor]:
+        return self._src, self._dst
+
+    def csc(self) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+        """(indptr, indices, eids): for dst v, in-neighbors are
+        indices[indptr[v]:indptr[v+1]] with original edge ids eids[...]."""
+        if self._csc is None:
+            self._csc = _coo_to_compressed(self._dst, self._src, self._num_nodes)
+        return self._csc
+
+    def csr(self) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+        """(indptr, indices, eids): for src u, out-neighbors are
+        indices[indptr[u]:indptr[u+1]] with original edge ids eids[...]."""
+        if self._csr is None:
+            self._csr = _coo_to_compressed(self._src, self._dst, self._num_nodes)
+        return self._csr
+
+    def in_degrees(self) -> torch.Tensor:
+        indptr, _, _ = self.csc()
+        return indptr[1:] - indptr[:-1]
+
+    def out_degrees(self) -> torch.Tensor:
+        indptr, _, _ = self.csr()
+        return indptr[1:] - indptr[:-1]
+
+    def to(self, device) -> "Graph":
+        g = Graph(self._src.to(device), self._dst.to(device), self._num_nodes)
+        if self._csc is not None:
+            g._csc = tuple(t.to(device) for t in self._csc)
+        if self._csr is not None:
+            g._csr = tuple(t.to(device) for t in self._csr)
+        g.ndata = {k: v.to(device) for k, v in self.ndata.items()}
+        g.edata = {k: v.to(device) for k, v in self.edata.items()}
+        return g
+
+    def add_self_loops(self) -> "Graph":
+        """Return a new graph with self-loop edges appended (GCN convention)."""
+        n = self._num_nodes
+        loop = torch.arange(n, dtype=torch.int64, device=self.device)
+        g = Graph(torch.cat([self._src, loop]), torch.cat([self._dst, loop]), n)
+        g.ndata = dict(self.ndata)
+        return g
+
+    def reverse(self) -> "Graph":
+        g = Graph(self._dst, self._src, self._num_nodes)
+        g.ndata = dict(self.ndata)
+        return g
+
+
+class Block:
+    """Bipartite message-flow graph (MFG) produced by neighbor sampling.
+
+    src node ids are block-local in [0, num_src); dst nodes are the first
+    ``num_dst`` src nodes. ``srcdata_nids`` maps block-local src index ->
+    parent-graph node id.
+    """
+
+    def __init__(
+        self,
+        indptr: torch.Tensor,
+        indices: torch.Tensor,
+        num_src: int,
+        num_dst: int,
+        srcdata_nids: Optional[torch.Tensor] = None,
+        eids: Optional[torch.Tensor] = None,
+    ):
+        assert indptr.numel() == num_dst + 1
+        self.csc_indptr = indptr
+        self.csc_indices = indices
+        self._num_src = int(num_src)
+        self._num_dst = int(num_dst)
+        self.srcdata_nids = srcdata_nids
+        self.csc_eids = eids
+        self._csr: Optional[Tuple[torch.Tensor, torch.Tensor, torch.Tensor]] = None
+        self.edata = {}
+
+    @property
+    def num_src_nodes(self) -> int:
+        return self._num_src
+
+    @property
+    def num_dst_nodes(self) -> int:
+        return self._num_dst
+
+    @property
+    def num_edges(self) -> int:
+        return self.csc_indices.numel()
+
+    @property
+    def device(self) -> torch.device:
+        return self.csc_indices.device
+
+    def csc(self) -> Tuple[torch.Tensor, torch.Tensor, Optional[torch.Tensor]]:
+        return self.csc_indptr, self.csc_indices, self.csc_eids
+
+    def csr(self) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+        """Transposed structure over src nodes (for backward SpMM)."""
+        if self._csr is None:
+            # dst index of each csc position
+            dst = torch.repeat_interleave(
+                torch.arange(self._num_dst, device=self.device),
+                self.csc_indptr[1:] - self.csc_indptr[:-1],
+            )
+            self._csr = _coo_to_compressed(self.csc_indices, dst, self._num_src)
+        return self._csr
+
+    def in_degrees(self) -> torch.Tensor:
+        return self.csc_indptr[1:] - self.csc_indptr[:-1]
+
+    def to(self, device) -> "Block":
+        b = Block(
+            self.csc_indptr.to(device),
+            self.csc_indices.to(device),
+            self._num_src,
+            self._num_dst,
+            None if self.srcdata_nids is None else self.srcdata_nids.to(device),
+            None if self.csc_eids is None else self.csc_eids.to(device),
+        )
+        if self._csr is not None:
+            b._csr = tuple(t.to(device) for t in self._csr)
+        b.edata = {k: v.to(device) for k, v in self.edata.items()}
+        return b
+
+
+def batch_graphs(graphs: Sequence[Graph]) -> Tuple[Graph, torch.Tensor]:
+    """Disjoint union of graphs (for graph classification); returns the batched
+    graph and ``batch_num_nodes`` (nodes per component, for segment readout).
+
+    Reference behavior: dgl.batch + dgl.mean_nodes in
+    /root/reference/examples/graph_classification/code/5_graph_classification.py.
+    """
+    srcs: List[torch.Tensor] = []
+    dsts: List[torch.Tensor] = []
+    sizes = []
+    off = 0
+    for g in graphs:
+        s, d = g.edges()
+        srcs.append(s + off)
+        dsts.append(d + off)
+        sizes.append(g.num_nodes)
+        off += g.num_nodes
+    bg = Graph(torch.cat(srcs), torch.cat(dsts), off)
+    return bg, torch.tensor(sizes, dtype=torch.int64)

@@ -1,0 +1,31 @@
+from . import backend
+from .spmm import gspmm, spmm_raw
+from .sddmm import sddmm_dot, edge_softmax, edge_softmax_csc
+from .segment import segment_reduce, mean_nodes
+from .sampling import (
+    sample_neighbors,
+    to_block,
+    NeighborSampler,
+    CompactionWorkspace,
+)
+from .kge import get_score_func, kge_loss, SCORE_FUNCS
+from .adagrad import sparse_adagrad_update
+
+__all__ = [
+    "backend",
+    "gspmm",
+    "spmm_raw",
+    "sddmm_dot",
+    "edge_softmax",
+    "edge_softmax_csc",
+    "segment_reduce",
+    "mean_nodes",
+    "sample_neighbors",
+    "to_block",
+    "NeighborSampler",
+    "CompactionWorkspace",
+    "get_score_func",
+    "kge_loss",
+    "SCORE_FUNCS",
+    "sparse_adagrad_update",
+]

@@ -1,0 +1,79 @@
+import torch
+
+from dgl_operator_amd.graph import Graph, Block, batch_graphs, rmat_graph
+
+
+def small_graph():
+    #  0 -> 1, 0 -> 2, 1 -> 2, 2 -> 0, 3 -> 2
+    src = torch.tensor([0, 0, 1, 2, 3])
+    dst = torch.tensor([1, 2, 2, 0, 2])
+    return Graph(src, dst, 4)
+
+
+def test_csc_structure():
+    g = small_graph()
+    indptr, indices, eids = g.csc()
+    assert indptr.tolist() == [0, 1, 2, 5, 5]
+    assert sorted(indices[1:2].tolist()) == [0]
+    assert sorted(indices[2:5].tolist()) == [0, 1, 3]
+    # eids map back to COO edges
+    src, dst = g.edges()
+    assert torch.equal(src[eids], indices)
+
+
+def test_csr_structure():
+    g = small_graph()
+    indptr, indices, eids = g.csr()
+    assert indptr.tolist() == [0, 2, 3, 4, 5]
+    assert sorted(indices[0:2].tolist()) == [1, 2]
+    src, dst = g.edges()
+    assert torch.equal(dst[eids], indices)
+
+
+def test_degrees():
+    g = small_graph()
+    assert g.in_degrees().tolist() == [1, 1, 3, 0]
+    assert g.out_degrees().tolist() == [2, 1, 1, 1]
+
+
+def test_self_loops_and_reverse():
+    g = small_graph().add_self_loops()
+    assert g.num_edges == 9
+    assert g.in_degrees().tolist() == [2, 2, 4, 1]
+    r = small_graph().reverse()
+    assert r.in_degrees().tolist() == [2, 1, 1, 1]
+
+
+def test_batch_graphs():
+    g1 = small_graph()
+    g2 = Graph(torch.tensor([0]), torch.tensor([1]), 2)
+    bg, sizes = batch_graphs([g1, g2])
+    assert bg.num_nodes == 6
+    assert bg.num_edges == 6
+    assert sizes.tolist() == [4, 2]
+    src, dst = bg.edges()
+    assert src[-1].item() == 4 and dst[-1].item() == 5
+
+
+def test_rmat_shape_and_determinism():
+    g1 = rmat_graph(1000, 5000, num_feats=8, num_classes=3, seed=7)
+    g2 = rmat_graph(1000, 5000, num_feats=8, num_classes=3, seed=7)
+    assert g1.num_nodes == 1000
+    assert g1.num_edges <= 5000  # self loops dropped
+    assert g1.num_edges > 4000
+    s1, d1 = g1.edges()
+    s2, d2 = g2.edges()
+    assert torch.equal(s1, s2) and torch.equal(d1, d2)
+    assert torch.equal(g1.ndata["feat"], g2.ndata["feat"])
+    # power-law-ish: max degree far above mean
+    assert g1.in_degrees().max().item() > 5 * g1.in_degrees().float().mean().item()
+
+
+def test_block_csr_transpose():
+    # block: 3 dst, 5 src; dst0 <- {0,3}, dst1 <- {1,4}, dst2 <- {2}
+    indptr = torch.tensor([0, 2, 4, 5])
+    indices = torch.tensor([0, 3, 1, 4, 2])
+    b = Block(indptr, indices, num_src=5, num_dst=3)
+    rindptr, rindices, reids = b.csr()
+    assert rindptr.tolist() == [0, 1, 2, 3, 4, 5]
+    assert rindices.tolist() == [0, 1, 2, 0, 1]
