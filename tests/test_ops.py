@@ -205,19 +205,20 @@ def test_kge_scores_shapes_and_consistency():
     tails = torch.randn(B, D)
     for name in ["TransE_l1", "TransE_l2", "DistMult", "ComplEx", "RotatE"]:
         fn = get_score_func(name, gamma=10.0)
-        rel = torch.randn(B, D // 2 if name == "RotatE" else D)
+        Dr = D // 2 if name == "RotatE" else D
+        rel = torch.randn(B, Dr)
         pos = fn.edge(heads, rel, tails)
         assert pos.shape == (B,)
         negs = torch.randn(C, NEG, D)
         ns = fn.neg(
-            heads.view(C, chunk, D), rel.view(C, chunk, D), negs, neg_head=False
+            heads.view(C, chunk, D), rel.view(C, chunk, Dr), negs, neg_head=False
         )
         assert ns.shape == (C, chunk, NEG)
         # consistency: neg score for entity j must equal edge score with tail j
         j = 2
         manual = fn.edge(
             heads.view(C, chunk, D)[0],
-            rel.view(C, chunk, D)[0],
+            rel.view(C, chunk, Dr)[0],
             negs[0, j].expand(chunk, D),
         )
         assert torch.allclose(ns[0, :, j], manual, atol=1e-4)
@@ -229,13 +230,14 @@ def test_kge_neg_head_consistency():
     tails = torch.randn(B, D)
     for name in ["TransE_l2", "ComplEx", "RotatE", "DistMult"]:
         fn = get_score_func(name, gamma=10.0)
-        rel = torch.randn(B, D // 2 if name == "RotatE" else D)
+        Dr = D // 2 if name == "RotatE" else D
+        rel = torch.randn(B, Dr)
         negs = torch.randn(C, NEG, D)
-        ns = fn.neg(tails.view(C, chunk, D), rel.view(C, chunk, D), negs, neg_head=True)
+        ns = fn.neg(tails.view(C, chunk, D), rel.view(C, chunk, Dr), negs, neg_head=True)
         j = 1
         manual = fn.edge(
             negs[0, j].expand(chunk, D),
-            rel.view(C, chunk, D)[0],
+            rel.view(C, chunk, Dr)[0],
             tails.view(C, chunk, D)[0],
         )
         assert torch.allclose(ns[0, :, j], manual, atol=1e-4), name
