@@ -1,0 +1,184 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: distributed GraphSAGE minibatch training on a synthetic
+R-MAT graph of ogbn-products shape (BASELINE.json metric: edges/sec, whole
+node).
+
+Reference config anchors (SURVEY.md §6 / BASELINE.md): 2-layer GraphSAGE
+hidden 16, batch 1000 seeds/rank, fanout [10,25], Adam lr 3e-3, dropout 0.5 —
+/root/reference/examples/GraphSAGE_dist/code/train_dist.py defaults; graph =
+2,449,029 nodes / 61,859,140 edges / 100 feats / 47 classes (ogbn-products
+shape), synthetic R-MAT with random-init weights (no dataset network access).
+
+Run (driver contract):
+  python bench.py --gpus 1 --steps K --warmup W
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+One process per GPU over RCCL; each rank owns one range partition
+(weak scaling: per-rank batch fixed at 1000 seeds). The timed step includes
+GPU neighbor sampling, alltoallv halo feature pull, forward, backward,
+gradient all-reduce and the Adam update. Metric value = SUM over ranks of
+message-passing edges in the sampled blocks / elapsed (max over ranks).
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+import torch.nn.functional as F
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--nodes", type=int, default=2_449_029)
+    p.add_argument("--edges", type=int, default=61_859_140)
+    p.add_argument("--feat", type=int, default=100)
+    p.add_argument("--classes", type=int, default=47)
+    p.add_argument("--hidden", type=int, default=16)
+    p.add_argument("--layers", type=int, default=2)
+    p.add_argument("--batch", type=int, default=1000)
+    p.add_argument("--fanout", type=str, default="10,25")
+    p.add_argument("--lr", type=float, default=3e-3)
+    p.add_argument("--dropout", type=float, default=0.5)
+    p.add_argument("--partition", type=str, default="range",
+                   help="range|ldg partition strategy for multi-rank runs")
+    p.add_argument("--device", type=str, default=None)
+    return p.parse_args()
+
+
+def flat_allreduce_grads(model):
+    if not dist.is_initialized():
+        return
+    grads = [p.grad for p in model.parameters() if p.grad is not None]
+    flat = torch.cat([g.reshape(-1) for g in grads])
+    dist.all_reduce(flat)
+    flat /= dist.get_world_size()
+    off = 0
+    for g in grads:
+        n = g.numel()
+        g.copy_(flat[off : off + n].view_as(g))
+        off += n
+
+
+def main():
+    args = parse_args()
+    from dgl_operator_amd.distributed import DistGraph, PartitionBook, comm
+    from dgl_operator_amd.graph import rmat_graph
+    from dgl_operator_amd.models import GraphSAGE
+
+    rank, ws = comm.init_from_env()
+    if args.device:
+        device = torch.device(args.device)
+    elif torch.cuda.is_available():
+        device = torch.device(f"cuda:{int(os.environ.get('LOCAL_RANK', 0))}")
+        torch.cuda.set_device(device)
+    else:
+        device = torch.device("cpu")
+
+    fanouts = [int(x) for x in args.fanout.split(",")]
+
+    # Every rank generates the same graph deterministically, then shards it.
+    g = rmat_graph(
+        args.nodes, args.edges, num_feats=args.feat, num_classes=args.classes,
+        seed=0, device=device,
+    )
+    n = g.num_nodes
+    bounds = [n * p // ws for p in range(ws + 1)]
+    book = PartitionBook(bounds, device=device)
+    dg = DistGraph.from_full_graph(g, book, rank)
+    # free the full graph copies we no longer need (features stay sharded)
+    del g
+
+    model = GraphSAGE(args.feat, args.hidden, args.classes,
+                      n_layers=args.layers, dropout=args.dropout).to(device)
+    # identical init on every rank (same torch seed), so no broadcast needed;
+    # make it explicit anyway for robustness
+    if ws > 1:
+        for p in model.parameters():
+            dist.broadcast(p.data, src=0)
+    opt = torch.optim.Adam(model.parameters(), lr=args.lr)
+
+    seed_gen = torch.Generator(device=device)
+    seed_gen.manual_seed(12345 + rank)
+    lo, hi = dg.lo, dg.hi
+
+    def one_step(step: int) -> int:
+        seeds = torch.randint(lo, hi, (args.batch,), generator=seed_gen,
+                              device=device)
+        seeds = torch.unique(seeds)
+        input_nodes, output_nodes, blocks = dg.sample_blocks(
+            seeds, fanouts, seed=step + 1
+        )
+        x = dg.pull("feat", input_nodes)
+        y = dg.pull("label", output_nodes)
+        logits = model(blocks, x)
+        loss = F.cross_entropy(logits, y)
+        opt.zero_grad(set_to_none=False)
+        loss.backward()
+        flat_allreduce_grads(model)
+        opt.step()
+        return sum(b.num_edges for b in blocks)
+
+    # warmup
+    for s in range(args.warmup):
+        one_step(s)
+
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    comm.barrier()
+    t0 = time.perf_counter()
+    edges = 0
+    for s in range(args.warmup, args.warmup + args.steps):
+        edges += one_step(s)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    comm.barrier()
+    elapsed = time.perf_counter() - t0
+
+    t = torch.tensor([elapsed], dtype=torch.float64)
+    e = torch.tensor([edges], dtype=torch.float64)
+    if ws > 1:
+        # max elapsed over ranks; sum of processed edges over ranks
+        te = t.to(device) if dist.get_backend() == "nccl" else t
+        ee = e.to(device) if dist.get_backend() == "nccl" else e
+        dist.all_reduce(te, op=dist.ReduceOp.MAX)
+        dist.all_reduce(ee, op=dist.ReduceOp.SUM)
+        elapsed = float(te.cpu()[0])
+        edges = float(ee.cpu()[0])
+
+    if rank == 0:
+        value = edges / elapsed
+        print(json.dumps({
+            "metric": "edges/sec (whole node) GraphSAGE ogbn-products",
+            "value": value,
+            "unit": "edges/s",
+            "n_gpus": ws,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic R-MAT, ogbn-products shape (2.45M nodes / 61.9M edges / 100 feats / 47 classes), random-init weights",
+            "config": {
+                "model": f"GraphSAGE L{args.layers} h{args.hidden} fanout[{args.fanout}] dropout{args.dropout}",
+                "global_batch": args.batch * ws,
+                "seq_len": None,
+                "parallelism": f"dp{ws} (graph-partition data parallel, alltoallv halo pulls)",
+            },
+        }))
+
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

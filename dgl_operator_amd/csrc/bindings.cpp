@@ -1,0 +1,40 @@
+// pybind bindings for the dgl_operator_amd native extension (_C.so).
+
+#include <torch/extension.h>
+
+namespace doa {
+
+at::Tensor spmm(at::Tensor indptr, at::Tensor indices, at::Tensor feat,
+                c10::optional<at::Tensor> eweight, bool mean);
+at::Tensor sddmm_dot(at::Tensor src, at::Tensor dst, at::Tensor feat_u,
+                     at::Tensor feat_v);
+at::Tensor edge_softmax_fwd(at::Tensor indptr, at::Tensor scores);
+at::Tensor edge_softmax_bwd(at::Tensor indptr, at::Tensor out,
+                            at::Tensor grad_out);
+at::Tensor segment_reduce(at::Tensor offsets, at::Tensor feat, bool mean);
+std::tuple<at::Tensor, at::Tensor> sample_neighbors(at::Tensor indptr,
+                                                    at::Tensor indices,
+                                                    at::Tensor seeds,
+                                                    int64_t fanout, bool replace,
+                                                    int64_t seed);
+void sparse_adagrad(at::Tensor emb, at::Tensor state, at::Tensor ids,
+                    at::Tensor grad, double lr, double eps);
+at::Tensor ldg_partition(at::Tensor indptr, at::Tensor indices,
+                         at::Tensor cindptr, at::Tensor cindices,
+                         int64_t num_parts);
+
+}  // namespace doa
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "dgl_operator_amd native HIP/CDNA4 kernels (gfx950)";
+  m.def("spmm", &doa::spmm, "generalized SpMM (copy_u/u_mul_e x sum/mean)",
+        py::arg("indptr"), py::arg("indices"), py::arg("feat"),
+        py::arg("eweight") = py::none(), py::arg("mean") = false);
+  m.def("sddmm_dot", &doa::sddmm_dot, "per-edge u dot v");
+  m.def("edge_softmax_fwd", &doa::edge_softmax_fwd);
+  m.def("edge_softmax_bwd", &doa::edge_softmax_bwd);
+  m.def("segment_reduce", &doa::segment_reduce);
+  m.def("sample_neighbors", &doa::sample_neighbors);
+  m.def("sparse_adagrad", &doa::sparse_adagrad);
+  m.def("ldg_partition", &doa::ldg_partition);
+}
