@@ -1,0 +1,67 @@
+"""watcher-loop equivalent: block until the pods named in a hostfile-format
+watch file reach Running (mode=ready) or Succeeded (mode=finished).
+
+Reference: /root/reference/watcher-loop/app/server.go:38-120 +
+controllers/controller.go:121-254 — pod names come from column 3 of the watch
+file, entries ending in "launcher" are skipped, a 500 ms ticker exits when the
+watched set drains. This implementation polls the Cluster abstraction (the
+informer machinery is a k8s-specific optimization; the observable contract —
+block-until-state — is identical).
+"""
+from __future__ import annotations
+
+import time
+from typing import Iterable, List, Optional
+
+from .api import PodPhase
+from .cluster import Cluster
+
+
+def parse_watchfile(text: str) -> List[str]:
+    """Hostfile format: `ip port podname slots=N`; returns pod names, skipping
+    any *launcher entry (watcher-loop/app/server.go:108-120)."""
+    names = []
+    for line in text.splitlines():
+        parts = line.split()
+        if len(parts) < 3:
+            continue
+        name = parts[2]
+        if name.endswith("launcher"):
+            continue
+        names.append(name)
+    return names
+
+
+def watch(
+    cluster: Cluster,
+    namespace: str,
+    pod_names: Iterable[str],
+    mode: str = "ready",
+    poll_interval: float = 0.5,
+    timeout: Optional[float] = None,
+) -> bool:
+    """Block until every pod satisfies the mode predicate. Returns True on
+    success, False on timeout."""
+    assert mode in ("ready", "finished")
+    watched = set(pod_names)
+    deadline = time.time() + timeout if timeout else None
+    while watched:
+        for name in list(watched):
+            p = cluster.get_pod(namespace, name)
+            if p is None:
+                continue
+            if mode == "ready" and p.is_real_running():
+                watched.discard(name)
+            elif mode == "finished" and p.phase == PodPhase.SUCCEEDED:
+                watched.discard(name)
+        if not watched:
+            return True
+        if deadline and time.time() > deadline:
+            return False
+        time.sleep(poll_interval)
+    return True
+
+
+def watch_file(cluster: Cluster, namespace: str, watchfile_text: str,
+               mode: str, **kw) -> bool:
+    return watch(cluster, namespace, parse_watchfile(watchfile_text), mode, **kw)

@@ -1,0 +1,244 @@
+"""Control-plane tests: the phase ladder the reference asserts in its envtest
+suite (/root/reference/controllers/dgljob_controller_test.go:131-214) —
+which never actually runs there because envtest has no kubelet. Here the
+FakeCluster's driveable kubelet makes the ladder testable:
+Pending -> Partitioning -> Partitioned -> Training -> Completed.
+"""
+import pytest
+
+from dgl_operator_amd.operator_plane import (
+    CleanPodPolicy,
+    DGLJobReconciler,
+    FakeCluster,
+    JobPhase,
+    PartitionMode,
+    PodPhase,
+    ReplicaType,
+    job_from_manifest,
+    watcher,
+)
+
+GRAPHSAGE_YAML = """
+apiVersion: qihoo.net/v1alpha1
+kind: DGLJob
+metadata:
+  name: graphsage-dist
+  namespace: default
+spec:
+  partitionMode: DGL-API
+  cleanPodPolicy: Running
+  slotsPerWorker: 1
+  dglReplicaSpecs:
+    Launcher:
+      replicas: 1
+      template:
+        spec:
+          containers:
+          - name: dgl-launcher
+            image: dgl-operator-amd:worker
+            command: ["dglrun"]
+            args: ["--graph-name", "ogbn-products", "--partition-entry-point",
+                   "code/load_and_partition_graph.py", "--num-partitions", "2",
+                   "--train-entry-point", "code/train_dist.py"]
+    Worker:
+      replicas: 2
+      template:
+        spec:
+          containers:
+          - name: dgl-worker
+            image: dgl-operator-amd:worker
+"""
+
+
+def make_job():
+    return job_from_manifest(GRAPHSAGE_YAML)
+
+
+def test_manifest_parsing():
+    job = make_job()
+    assert job.name == "graphsage-dist"
+    assert job.spec.partition_mode == PartitionMode.DGL_API
+    assert job.spec.clean_pod_policy == CleanPodPolicy.RUNNING
+    assert job.num_workers() == 2
+    assert job.spec.replica_specs[ReplicaType.LAUNCHER].template["spec"][
+        "containers"
+    ][0]["command"] == ["dglrun"]
+
+
+def test_phase_ladder_full():
+    c = FakeCluster()
+    r = DGLJobReconciler(c)
+    job = make_job()
+
+    # 1. first reconcile: launcher + partitioner created, workers gated
+    r.reconcile(job)
+    assert c.get_pod("default", "graphsage-dist-launcher") is not None
+    assert c.get_pod("default", "graphsage-dist-partitioner") is not None
+    assert c.get_pod("default", "graphsage-dist-worker-0") is None
+    assert job.status.phase in (JobPhase.PENDING, JobPhase.STARTING)
+    assert job.status.start_time is not None
+
+    # 2. partitioner runs -> Partitioning
+    c.set_pod_phase("default", "graphsage-dist-partitioner", PodPhase.RUNNING)
+    r.reconcile(job)
+    assert job.status.phase == JobPhase.PARTITIONING
+    assert job.status.replica_statuses[ReplicaType.PARTITIONER].active == 1
+
+    # 3. partitioner succeeds -> Partitioned; workers get created
+    c.set_pod_phase("default", "graphsage-dist-partitioner", PodPhase.SUCCEEDED)
+    r.reconcile(job)
+    assert job.status.phase == JobPhase.PARTITIONED
+    assert c.get_pod("default", "graphsage-dist-worker-0") is not None
+    assert c.get_pod("default", "graphsage-dist-worker-1") is not None
+    # headless service per worker
+    svc = c.get_service("default", "graphsage-dist-worker-0")
+    assert svc is not None and svc.cluster_ip is None
+    assert svc.selector == {"dgl-replica-name": "graphsage-dist-worker-0"}
+
+    # 4. workers + launcher run -> Training; hostfile has worker IPs
+    c.run_all_pending()
+    r.reconcile(job)
+    assert job.status.phase == JobPhase.TRAINING
+    ws = job.status.replica_statuses[ReplicaType.WORKER]
+    assert ws.active == 2 and ws.ready == "2/2"
+    cm = c.get_configmap("default", "graphsage-dist-config")
+    lines = cm.data["hostfile"].strip().splitlines()
+    assert len(lines) == 2
+    assert lines[0].split()[2] == "graphsage-dist-worker-0"
+    assert lines[0].split()[1] == "30050"
+    assert lines[0].split()[3] == "slots=1"
+    assert "kubexec.sh" in cm.data
+
+    # 5. launcher succeeds -> Completed
+    c.set_pod_phase("default", "graphsage-dist-launcher", PodPhase.SUCCEEDED)
+    r.reconcile(job)
+    assert job.status.phase == JobPhase.COMPLETED
+    assert job.status.completion_time is not None
+    ls = job.status.replica_statuses[ReplicaType.LAUNCHER]
+    assert ls.succeeded == 1
+
+
+def test_failed_pod_fails_job():
+    c = FakeCluster()
+    r = DGLJobReconciler(c)
+    job = make_job()
+    r.reconcile(job)
+    c.set_pod_phase("default", "graphsage-dist-partitioner", PodPhase.FAILED)
+    r.reconcile(job)
+    assert job.status.phase == JobPhase.FAILED
+
+
+def test_clean_pod_policy_running_keeps_finished():
+    c = FakeCluster()
+    r = DGLJobReconciler(c)
+    job = make_job()
+    r.reconcile(job)
+    c.set_pod_phase("default", "graphsage-dist-partitioner", PodPhase.SUCCEEDED)
+    r.reconcile(job)
+    c.run_all_pending()
+    r.reconcile(job)
+    c.set_pod_phase("default", "graphsage-dist-worker-0", PodPhase.SUCCEEDED)
+    c.set_pod_phase("default", "graphsage-dist-launcher", PodPhase.SUCCEEDED)
+    r.reconcile(job)
+    assert job.status.phase == JobPhase.COMPLETED
+    # next reconcile applies cleanPodPolicy Running: running pods deleted,
+    # finished ones kept (dgljob_controller.go:776-783)
+    r.reconcile(job)
+    assert c.get_pod("default", "graphsage-dist-worker-0") is not None  # Succeeded
+    assert c.get_pod("default", "graphsage-dist-worker-1") is None  # was Running
+    assert c.get_pod("default", "graphsage-dist-launcher") is not None
+
+
+def test_skip_mode_reaches_completed():
+    yaml_skip = GRAPHSAGE_YAML.replace("DGL-API", "Skip")
+    job = job_from_manifest(yaml_skip)
+    c = FakeCluster()
+    r = DGLJobReconciler(c)
+    r.reconcile(job)
+    # no partitioner pod in Skip mode; launcher env has Launcher_Workload
+    assert c.get_pod("default", "graphsage-dist-partitioner") is None
+    lp = c.get_pod("default", "graphsage-dist-launcher")
+    assert lp.spec["env"]["DGL_OPERATOR_PHASE_ENV"] == "Launcher_Workload"
+    c.run_all_pending()
+    r.reconcile(job)
+    assert job.status.phase == JobPhase.TRAINING
+    c.set_pod_phase("default", "graphsage-dist-launcher", PodPhase.SUCCEEDED)
+    r.reconcile(job)
+    assert job.status.phase == JobPhase.COMPLETED
+
+
+def test_parmetis_mode_runs_partitioner():
+    yaml_pm = GRAPHSAGE_YAML.replace("DGL-API", "ParMETIS")
+    job = job_from_manifest(yaml_pm)
+    c = FakeCluster()
+    r = DGLJobReconciler(c)
+    r.reconcile(job)
+    assert c.get_pod("default", "graphsage-dist-partitioner") is not None
+
+
+def test_rbac_scoping():
+    c = FakeCluster()
+    r = DGLJobReconciler(c)
+    job = make_job()
+    r.reconcile(job)
+    role = c.get_rbac("default", "Role", "graphsage-dist-launcher")
+    exec_rule = [ru for ru in role.rules if "pods/exec" in ru["resources"]][0]
+    assert set(exec_rule["resourceNames"]) == {
+        "graphsage-dist-worker-0",
+        "graphsage-dist-worker-1",
+    }
+    prole = c.get_rbac("default", "Role", "graphsage-dist-partitioner")
+    exec_rule = [ru for ru in prole.rules if "pods/exec" in ru["resources"]][0]
+    assert exec_rule["resourceNames"] == ["graphsage-dist-launcher"]
+
+
+def test_partitioner_inherits_launcher_command():
+    c = FakeCluster()
+    r = DGLJobReconciler(c)
+    job = make_job()
+    r.reconcile(job)
+    p = c.get_pod("default", "graphsage-dist-partitioner")
+    assert p.spec["command"] == ["dglrun"]
+    assert p.spec["env"]["DGL_OPERATOR_PHASE_ENV"] == "Partitioner"
+    # worker ports 30050-30069 (HOST_PORT_NUM=20)
+    w = None
+    c.set_pod_phase("default", "graphsage-dist-partitioner", PodPhase.SUCCEEDED)
+    r.reconcile(job)
+    w = c.get_pod("default", "graphsage-dist-worker-0")
+    assert w.spec["ports"][0] == 30050 and len(w.spec["ports"]) == 20
+    assert w.spec["command"] == ["sleep", "365d"]
+
+
+def test_deletion_cleans_all():
+    c = FakeCluster()
+    r = DGLJobReconciler(c)
+    job = make_job()
+    r.reconcile(job)
+    c.run_all_pending()
+    job.deletion_timestamp = 1.0
+    r.reconcile(job)
+    assert c.list_pods("default", "graphsage-dist") == []
+
+
+def test_watcher_ready_and_finished():
+    c = FakeCluster()
+    r = DGLJobReconciler(c)
+    job = make_job()
+    r.reconcile(job)
+    c.set_pod_phase("default", "graphsage-dist-partitioner", PodPhase.SUCCEEDED)
+    r.reconcile(job)
+    c.run_all_pending()
+    r.reconcile(job)
+    cm = c.get_configmap("default", "graphsage-dist-config")
+    names = watcher.parse_watchfile(cm.data["hostfile"])
+    assert names == ["graphsage-dist-worker-0", "graphsage-dist-worker-1"]
+    assert watcher.watch(c, "default", names, "ready", timeout=2)
+    # finished-mode times out while workers still run
+    assert not watcher.watch(
+        c, "default", names, "finished", poll_interval=0.05, timeout=0.2
+    )
+    c.set_pod_phase("default", "graphsage-dist-worker-0", PodPhase.SUCCEEDED)
+    c.set_pod_phase("default", "graphsage-dist-worker-1", PodPhase.SUCCEEDED)
+    assert watcher.watch(c, "default", names, "finished", timeout=2)
+    # launcher entries are skipped
+    assert watcher.parse_watchfile(cm.data["leadfile"]) == []
