@@ -1,0 +1,134 @@
+"""Distributed KGE training — the dglke_dist_train equivalent.
+
+Entity AND relation embeddings live in ShardedEmbeddings (one shard per GPU
+over xGMI; the reference partitions relations too — hotfix/kvclient.py:56-72).
+Each step: pull rows, score + self-adversarial loss with autograd over the
+pulled rows, push grads to their owning shards where fused sparse Adagrad
+applies them.
+
+Reference call stack being replaced: SURVEY.md §3.4 (kvclient ->
+dist_train_test -> KVClient.pull/push -> KGEServer sparse Adagrad).
+The head/tail corruption alternates per step like the reference's
+NewBidirectionalOneShotIterator (hotfix/sampler.py:823-875).
+"""
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+from ..ops import get_score_func, kge_loss
+from ..ops.adagrad import sparse_adagrad_update
+from . import comm
+from .kvstore import ShardedEmbedding
+
+
+class DistKGEModel:
+    def __init__(
+        self,
+        num_entities: int,
+        num_relations: int,
+        hidden_dim: int,
+        score_func: str = "TransE_l2",
+        gamma: float = 12.0,
+        rank: int = 0,
+        world_size: int = 1,
+        device="cpu",
+        seed: int = 0,
+    ):
+        self.num_entities = num_entities
+        self.hidden_dim = hidden_dim
+        emb_init = (gamma + 2.0) / hidden_dim
+        self.score = get_score_func(score_func, gamma=gamma, emb_init=emb_init)
+        self.entities = ShardedEmbedding(
+            num_entities, hidden_dim, world_size, rank, device=device,
+            init_range=emb_init, seed=seed,
+        )
+        rel_dim = hidden_dim
+        if score_func == "RotatE":
+            rel_dim = hidden_dim // 2
+        elif score_func == "RESCAL":
+            rel_dim = hidden_dim * hidden_dim
+        elif score_func == "TransR":
+            rel_dim = hidden_dim * (1 + hidden_dim)
+        # relations sharded too, like the reference's relation partition
+        # (hotfix/kvclient.py:56-72 round-robin); ranges work equally here
+        self.relations = ShardedEmbedding(
+            num_relations, rel_dim, world_size, rank, device=device,
+            init_range=emb_init, seed=seed + 7,
+        )
+
+    def train_step(
+        self,
+        heads: torch.Tensor,
+        rels: torch.Tensor,
+        tails: torch.Tensor,
+        neg_entities: torch.Tensor,  # [num_chunk, neg]
+        chunk_size: int,
+        lr: float,
+        neg_head: bool = False,
+        adversarial_temperature: float = 1.0,
+    ) -> float:
+        B = heads.numel()
+        num_chunk, n_neg = neg_entities.shape
+        assert B == num_chunk * chunk_size
+        dim = self.hidden_dim
+        h = self.entities.pull(heads).requires_grad_(True)
+        t = self.entities.pull(tails).requires_grad_(True)
+        n = self.entities.pull(neg_entities.reshape(-1)).requires_grad_(True)
+        r = self.relations.pull(rels).requires_grad_(True)
+
+        pos = self.score.edge(h, r, t)
+        hc = (t if neg_head else h).view(num_chunk, chunk_size, dim)
+        rc = r.view(num_chunk, chunk_size, -1)
+        nc = n.view(num_chunk, n_neg, dim)
+        neg = self.score.neg(hc, rc, nc, neg_head=neg_head)
+        loss = kge_loss(pos, neg, adversarial_temperature)
+        loss.backward()
+
+        with torch.no_grad():
+            ent_ids = torch.cat([heads, tails, neg_entities.reshape(-1)])
+            ent_grad = torch.cat([h.grad, t.grad, n.grad])
+            self.entities.push_grad(ent_ids, ent_grad, lr)
+            self.relations.push_grad(rels, r.grad, lr)
+        return float(loss.detach())
+
+
+class KGEdgeSampler:
+    """Chunked negative edge sampler (K8). Yields
+    (heads, rels, tails, neg_entities, neg_head) with head/tail corruption
+    alternating per step. Negatives are uniform entity draws shared per chunk
+    (DGL-KE's chunked corruption shape: batch/chunk_size chunks x neg)."""
+
+    def __init__(
+        self,
+        triples: Tuple[torch.Tensor, torch.Tensor, torch.Tensor],
+        num_entities: int,
+        batch_size: int = 1024,
+        neg_sample_size: int = 256,
+        chunk_size: int = 64,
+        seed: int = 0,
+        device="cpu",
+    ):
+        self.h, self.r, self.t = (x.to(device) for x in triples)
+        self.num_entities = num_entities
+        self.batch_size = batch_size
+        self.neg = neg_sample_size
+        self.chunk = chunk_size
+        self.gen = torch.Generator(device=device)
+        self.gen.manual_seed(seed)
+        self.step = 0
+        self.device = device
+
+    def next_batch(self):
+        E = self.h.numel()
+        idx = torch.randint(0, E, (self.batch_size,), generator=self.gen,
+                            device=self.device)
+        num_chunk = self.batch_size // self.chunk
+        negs = torch.randint(
+            0, self.num_entities, (num_chunk, self.neg), generator=self.gen,
+            device=self.device,
+        )
+        neg_head = self.step % 2 == 1
+        self.step += 1
+        return self.h[idx], self.r[idx], self.t[idx], negs, neg_head

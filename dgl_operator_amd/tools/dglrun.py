@@ -1,0 +1,117 @@
+"""dglrun — the 5-phase in-container workflow driver.
+
+Reference: /root/reference/python/dglrun/exec/dglrun (bash, 238 LoC).
+Branches on DGL_OPERATOR_PHASE_ENV exactly like the reference:
+  Launcher_Workload -> run the training entry point only (Skip mode)
+  Partitioner       -> Phase 1 partition + Phase 2 deliver to launcher
+  (unset, launcher) -> Phase 3 dispatch + Phase 4 revise + Phase 5 train
+Each phase echoes its wall-clock time like the reference does
+(dglrun:117-238).
+"""
+from __future__ import annotations
+
+import argparse
+import os
+import subprocess
+import sys
+import time
+from contextlib import contextmanager
+
+from .dispatch import dispatch_partitions
+from .fabric import get_fabric
+from .hostfile import parse_hostfile, revise_for_dgl
+from . import launch as launch_mod
+
+
+@contextmanager
+def phase(name: str):
+    t0 = time.time()
+    print(f"[dglrun] ---- {name} start ----", flush=True)
+    yield
+    print(f"[dglrun] ---- {name} done in {time.time() - t0:.1f}s ----", flush=True)
+
+
+def _run(cmd: str, env=None):
+    full_env = dict(os.environ)
+    full_env.update(env or {})
+    rc = subprocess.call(cmd, shell=True, env=full_env)
+    if rc != 0:
+        raise SystemExit(f"[dglrun] command failed rc={rc}: {cmd}")
+
+
+def build_parser():
+    p = argparse.ArgumentParser(prog="dglrun")
+    p.add_argument("--graph-name", default="graph")
+    p.add_argument("--partition-entry-point", default="")
+    p.add_argument("--partition-entry-args", default="")
+    p.add_argument("--num-partitions", type=int, default=1)
+    p.add_argument("--train-entry-point", default="")
+    p.add_argument("--train-entry-args", default="")
+    p.add_argument("--workspace", default=os.environ.get("WORKSPACE",
+                                                         "/dgl_workspace"))
+    p.add_argument("--hostfile", default="/etc/dgl/hostfile")
+    p.add_argument("--leadfile", default="/etc/dgl/leadfile")
+    p.add_argument("--master-port", type=int, default=29400)
+    return p
+
+
+def run_partitioner(args):
+    dataset = os.path.join(args.workspace, "dataset")
+    os.makedirs(dataset, exist_ok=True)
+    with phase("Phase 1/5 partition"):
+        _run(
+            f"python {args.partition_entry_point} "
+            f"--graph-name {args.graph_name} "
+            f"--num-partitions {args.num_partitions} "
+            f"--output {dataset} {args.partition_entry_args}"
+        )
+    with phase("Phase 2/5 deliver"):
+        with open(args.leadfile) as f:
+            leads = parse_hostfile(f.read())
+        fabric = get_fabric()
+        for lead in leads:
+            # copy into the launcher's still-running watcher-loop-partitioner
+            # init container (reference trick, dgljob_controller.go:1129-1138)
+            fabric.copy(dataset, lead.pod, f"{args.workspace}/dataset",
+                        container="watcher-loop-partitioner")
+
+
+def run_launcher(args):
+    with open(args.hostfile) as f:
+        hosts = parse_hostfile(f.read())
+    dataset = os.path.join(args.workspace, "dataset")
+    with phase("Phase 3/5 dispatch"):
+        dispatch_partitions(dataset, args.graph_name, hosts,
+                            workspace=args.workspace)
+    with phase("Phase 4/5 revise hostfile"):
+        revised = revise_for_dgl(hosts)
+        launch_mod.exec_batch(
+            hosts,
+            f"sh -c 'mkdir -p {args.workspace} && printf %s \"{revised}\" "
+            f"> {args.workspace}/hostfile_revised'",
+        )
+    with phase("Phase 5/5 train"):
+        launch_mod.train(
+            hosts,
+            args.train_entry_point,
+            f"--graph-name {args.graph_name} "
+            f"--part-config {args.workspace}/workload/{args.graph_name}.json "
+            f"{args.train_entry_args}",
+            master_port=args.master_port,
+        )
+
+
+def main(argv=None):
+    args = build_parser().parse_args(argv)
+    mode = os.environ.get("DGL_OPERATOR_PHASE_ENV", "")
+    if mode == "Launcher_Workload":
+        with phase("workload (Skip mode)"):
+            _run(f"python {args.train_entry_point} {args.train_entry_args}")
+    elif mode == "Partitioner":
+        run_partitioner(args)
+    else:
+        run_launcher(args)
+
+
+if __name__ == "__main__":
+    main()

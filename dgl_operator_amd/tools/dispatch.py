@@ -1,0 +1,77 @@
+"""Partition dispatcher: runs on the launcher after the partitioner delivered
+its output; rewrites per-part paths into the workers' workload dir and ships
+each part to its worker over the fabric.
+
+Reference: /root/reference/python/dglrun/tools/dispatch.py:52-91 (one
+partition per worker is enforced there at launch.py:107-108 — same check
+here).
+"""
+from __future__ import annotations
+
+import json
+import os
+from typing import List, Optional
+
+from .fabric import Fabric, get_fabric
+from .hostfile import HostEntry, parse_hostfile
+
+WORKLOAD_DIR = "workload"
+
+
+def dispatch_partitions(
+    dataset_dir: str,
+    graph_name: str,
+    hosts: List[HostEntry],
+    fabric: Optional[Fabric] = None,
+    workspace: str = "/dgl_workspace",
+) -> dict:
+    """Rewrite graph.json paths for the worker side and copy part payloads."""
+    fabric = fabric or get_fabric()
+    meta_path = os.path.join(dataset_dir, f"{graph_name}.json")
+    with open(meta_path) as f:
+        meta = json.load(f)
+    num_parts = meta["num_parts"]
+    assert num_parts == len(hosts), (
+        f"one partition per worker required: {num_parts} parts vs "
+        f"{len(hosts)} workers"
+    )
+    revised = dict(meta)
+    for i in range(num_parts):
+        entry = dict(meta[f"part-{i}"])
+        for k, v in list(entry.items()):
+            entry[k] = os.path.join(WORKLOAD_DIR, f"part{i}", os.path.basename(v))
+        revised[f"part-{i}"] = entry
+    revised_path = os.path.join(dataset_dir, f"{graph_name}_revised.json")
+    with open(revised_path, "w") as f:
+        json.dump(revised, f, indent=2)
+
+    for i, host in enumerate(hosts):
+        part_src = os.path.join(dataset_dir, f"part{i}")
+        dst_dir = f"{workspace}/{WORKLOAD_DIR}"
+        fabric.exec(host.pod, f"mkdir -p {dst_dir}/part{i}")
+        for fname in os.listdir(part_src):
+            fabric.copy(
+                os.path.join(part_src, fname), host.pod,
+                f"{dst_dir}/part{i}/{fname}",
+            )
+        fabric.copy(revised_path, host.pod, f"{dst_dir}/{graph_name}.json")
+    return revised
+
+
+def main(argv=None):
+    import argparse
+
+    p = argparse.ArgumentParser()
+    p.add_argument("--dataset-dir", required=True)
+    p.add_argument("--graph-name", required=True)
+    p.add_argument("--hostfile", default="/etc/dgl/hostfile")
+    p.add_argument("--workspace", default="/dgl_workspace")
+    args = p.parse_args(argv)
+    with open(args.hostfile) as f:
+        hosts = parse_hostfile(f.read())
+    dispatch_partitions(args.dataset_dir, args.graph_name, hosts,
+                        workspace=args.workspace)
+
+
+if __name__ == "__main__":
+    main()

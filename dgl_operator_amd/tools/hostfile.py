@@ -1,0 +1,72 @@
+"""Hostfile parsing + ipconfig revision.
+
+Reference: /root/reference/python/dglrun/tools/revise_hostfile.py — the
+operator writes `ip port podname slots=N` per worker; trainers need
+`ip port` (DGL) or `ip port num_servers` (DGL-KE).
+"""
+from __future__ import annotations
+
+import os
+from dataclasses import dataclass
+from typing import List, Tuple
+
+
+@dataclass
+class HostEntry:
+    ip: str
+    port: int
+    pod: str
+    slots: int
+
+
+def parse_hostfile(text: str) -> List[HostEntry]:
+    out = []
+    for line in text.splitlines():
+        parts = line.split()
+        if len(parts) < 4:
+            continue
+        out.append(
+            HostEntry(
+                ip=parts[0],
+                port=int(parts[1]),
+                pod=parts[2],
+                slots=int(parts[3].split("=")[1]),
+            )
+        )
+    return out
+
+
+def revise_for_dgl(entries: List[HostEntry]) -> str:
+    return "".join(f"{e.ip} {e.port}\n" for e in entries)
+
+
+def revise_for_dglke(entries: List[HostEntry], num_servers: int = 1) -> str:
+    return "".join(f"{e.ip} {e.port} {num_servers}\n" for e in entries)
+
+
+def main(argv=None):
+    import argparse
+
+    p = argparse.ArgumentParser(description="revise hostfile into ipconfig")
+    p.add_argument("--hostfile", default="/etc/dgl/hostfile")
+    p.add_argument("--output", default=None)
+    p.add_argument("--format", choices=["dgl", "dglke"], default="dgl")
+    p.add_argument("--num-servers", type=int, default=1)
+    args = p.parse_args(argv)
+    with open(args.hostfile) as f:
+        entries = parse_hostfile(f.read())
+    text = (
+        revise_for_dgl(entries)
+        if args.format == "dgl"
+        else revise_for_dglke(entries, args.num_servers)
+    )
+    out = args.output or os.path.join(
+        os.environ.get("WORKSPACE", "."), "hostfile_revised"
+    )
+    with open(out, "w") as f:
+        f.write(text)
+    print(f"wrote {out} ({len(entries)} hosts)")
+
+
+if __name__ == "__main__":
+    main()
