@@ -39,7 +39,8 @@ def dispatch_partitions(
     for i in range(num_parts):
         entry = dict(meta[f"part-{i}"])
         for k, v in list(entry.items()):
-            entry[k] = os.path.join(WORKLOAD_DIR, f"part{i}", os.path.basename(v))
+            # relative to the revised json's own directory (workload/)
+            entry[k] = os.path.join(f"part{i}", os.path.basename(v))
         revised[f"part-{i}"] = entry
     revised_path = os.path.join(dataset_dir, f"{graph_name}_revised.json")
     with open(revised_path, "w") as f:

@@ -1,0 +1,132 @@
+#!/usr/bin/env python3
+"""Distributed GraphSAGE training over a partitioned graph — the workload the
+reference's GraphSAGE_dist example runs
+(/root/reference/examples/GraphSAGE_dist/code/train_dist.py), rebuilt on the
+MI355X-native stack: HIP sampler + SpMM, RCCL alltoallv feature pulls, one
+rank per GPU (no server/sampler processes).
+
+Launched by dglrun Phase 5 as:
+  python -m torch.distributed.run --nnodes=P --node-rank=i ... train_dist.py
+      --graph-name G --part-config /dgl_workspace/workload/G.json ...
+Each node loads its own partition (node_rank == partition id).
+"""
+from __future__ import annotations
+
+import os as _os
+import sys as _sys
+
+_sys.path.insert(0, _os.path.join(_os.path.dirname(_os.path.abspath(__file__)), "..", ".."))
+
+
+
+import argparse
+import os
+import time
+
+import torch
+import torch.distributed as dist
+import torch.nn.functional as F
+
+
+def parse_args():
+    p = argparse.ArgumentParser(description="GraphSAGE distributed trainer")
+    p.add_argument("--graph-name", type=str, default="graph")
+    p.add_argument("--part-config", type=str, required=True)
+    p.add_argument("--num-epochs", type=int, default=1)
+    p.add_argument("--num-hidden", type=int, default=16)
+    p.add_argument("--num-layers", type=int, default=2)
+    p.add_argument("--fan-out", type=str, default="10,25")
+    p.add_argument("--batch-size", type=int, default=1000)
+    p.add_argument("--lr", type=float, default=0.003)
+    p.add_argument("--dropout", type=float, default=0.5)
+    p.add_argument("--log-every", type=int, default=20)
+    p.add_argument("--eval-every", type=int, default=0)
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    from dgl_operator_amd.distributed import DistGraph, comm
+
+    rank, ws = comm.init_from_env()
+    if torch.cuda.is_available():
+        device = torch.device(f"cuda:{int(os.environ.get('LOCAL_RANK', 0))}")
+        torch.cuda.set_device(device)
+    else:
+        device = torch.device("cpu")
+
+    # partition id = node rank (one partition per worker pod, slots GPUs each;
+    # with slots=1 this is rank)
+    part_id = int(os.environ.get("GROUP_RANK", os.environ.get("RANK", rank)))
+    dg = DistGraph.from_partition(args.part_config, part_id, device=device)
+
+    from dgl_operator_amd.models import GraphSAGE
+
+    feat = dg.ndata["feat"]
+    in_feats = feat.shape[1]
+    labels = dg.ndata["label"]
+    n_classes = int(labels.max().item()) + 1
+    nc = torch.tensor([n_classes], device=device)
+    if ws > 1:
+        dist.all_reduce(nc, op=dist.ReduceOp.MAX)
+    n_classes = int(nc.item())
+
+    model = GraphSAGE(in_feats, args.num_hidden, n_classes,
+                      n_layers=args.num_layers, dropout=args.dropout).to(device)
+    if ws > 1:
+        for p in model.parameters():
+            dist.broadcast(p.data, src=0)
+    opt = torch.optim.Adam(model.parameters(), lr=args.lr)
+    fanouts = [int(x) for x in args.fan_out.split(",")]
+
+    train_mask = dg.ndata.get("train_mask")
+    owned = dg.owned_nodes()
+    train_nids = owned[train_mask.bool()] if train_mask is not None else owned
+
+    gen = torch.Generator(device=device)
+    gen.manual_seed(1234 + rank)
+    steps_per_epoch = max(1, train_nids.numel() // args.batch_size)
+
+    for epoch in range(args.num_epochs):
+        t_epoch = time.time()
+        for step in range(steps_per_epoch):
+            tic = time.time()
+            sel = torch.randint(0, train_nids.numel(), (args.batch_size,),
+                                generator=gen, device=device)
+            seeds = torch.unique(train_nids[sel])
+            inp, out_nodes, blocks = dg.sample_blocks(
+                seeds, fanouts, seed=epoch * 100000 + step
+            )
+            x = dg.pull("feat", inp)
+            y = dg.pull("label", out_nodes)
+            t_sample = time.time()
+            logits = model(blocks, x)
+            loss = F.cross_entropy(logits, y)
+            t_fwd = time.time()
+            opt.zero_grad()
+            loss.backward()
+            if ws > 1:
+                for p in model.parameters():
+                    if p.grad is not None:
+                        dist.all_reduce(p.grad)
+                        p.grad /= ws
+            opt.step()
+            t_bwd = time.time()
+            if step % args.log_every == 0 and rank == 0:
+                n_edges = sum(b.num_edges for b in blocks)
+                print(
+                    f"Epoch {epoch:03d} | Step {step:05d} | Loss {loss:.4f} | "
+                    f"Edges {n_edges} | sample {t_sample - tic:.3f}s "
+                    f"fwd {t_fwd - t_sample:.3f}s bwd+upd {t_bwd - t_fwd:.3f}s",
+                    flush=True,
+                )
+        if rank == 0:
+            print(f"Epoch {epoch:03d} time {time.time() - t_epoch:.2f}s",
+                  flush=True)
+
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

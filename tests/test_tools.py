@@ -68,7 +68,7 @@ def test_dispatch_over_local_fabric(tmp_path):
         assert os.path.exists(f"{pd}/w/workload/part{i}/graph.pt")
         assert os.path.exists(f"{pd}/w/workload/part{i}/node_feat.pt")
         meta = json.load(open(f"{pd}/w/workload/toy.json"))
-        assert meta[f"part-{i}"]["part_graph"] == f"workload/part{i}/graph.pt"
+        assert meta[f"part-{i}"]["part_graph"] == f"part{i}/graph.pt"
 
 
 def test_dispatch_rejects_mismatched_counts(tmp_path):
