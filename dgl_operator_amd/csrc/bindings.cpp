@@ -17,6 +17,9 @@ std::tuple<at::Tensor, at::Tensor> sample_neighbors(at::Tensor indptr,
                                                     at::Tensor seeds,
                                                     int64_t fanout, bool replace,
                                                     int64_t seed);
+std::tuple<at::Tensor, at::Tensor> compact_ids(at::Tensor table,
+                                               at::Tensor seeds,
+                                               at::Tensor neighbors);
 void sparse_adagrad(at::Tensor emb, at::Tensor state, at::Tensor ids,
                     at::Tensor grad, double lr, double eps);
 at::Tensor ldg_partition(at::Tensor indptr, at::Tensor indices,
@@ -35,6 +38,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("edge_softmax_bwd", &doa::edge_softmax_bwd);
   m.def("segment_reduce", &doa::segment_reduce);
   m.def("sample_neighbors", &doa::sample_neighbors);
+  m.def("compact_ids", &doa::compact_ids);
   m.def("sparse_adagrad", &doa::sparse_adagrad);
   m.def("ldg_partition", &doa::ldg_partition);
 }

@@ -97,3 +97,112 @@ std::tuple<at::Tensor, at::Tensor> sample_neighbors(at::Tensor indptr,
 }
 
 }  // namespace doa
+
+namespace doa {
+
+// ---------------------------------------------------------------------------
+// Block compaction (the dgl.to_block relabel step, fused):
+// given a -1-filled translation table, seeds (assumed unique) and the sampled
+// neighbor list, assign block-local ids — seeds take [0, n_seed), unseen
+// neighbors claim ids n_seed + atomic counter — translate the neighbor list,
+// and hand back srcdata_nids. Ordering of new ids is nondeterministic
+// (atomic claim order), which is fine: features are gathered by nid.
+// Replaces a sort-based torch.unique + scatter chain (5+ kernels) with 4
+// tiny kernels and no sort.
+// ---------------------------------------------------------------------------
+
+__global__ void compact_seed_kernel(int64_t* __restrict__ table,
+                                    const int64_t* __restrict__ seeds,
+                                    int64_t* __restrict__ srcdata,
+                                    int64_t n_seed) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n_seed;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    table[seeds[i]] = i;
+    srcdata[i] = seeds[i];
+  }
+}
+
+__global__ void compact_claim_kernel(int64_t* __restrict__ table,
+                                     const int64_t* __restrict__ nbrs,
+                                     int64_t* __restrict__ srcdata,
+                                     unsigned long long* __restrict__ counter,
+                                     int64_t n_seed, int64_t n_nbr) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n_nbr;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t v = nbrs[i];
+    // claim unseen nodes: CAS -1 -> -2, winner assigns the real id
+    const long long old = atomicCAS(
+        reinterpret_cast<unsigned long long*>(&table[v]),
+        (unsigned long long)(-1LL), (unsigned long long)(-2LL));
+    if ((long long)old == -1LL) {
+      const int64_t id = n_seed + (int64_t)atomicAdd(counter, 1ull);
+      srcdata[id] = v;
+      atomicExch(reinterpret_cast<unsigned long long*>(&table[v]),
+                 (unsigned long long)id);
+    }
+  }
+}
+
+__global__ void compact_translate_kernel(const int64_t* __restrict__ table,
+                                         const int64_t* __restrict__ nbrs,
+                                         int64_t* __restrict__ out,
+                                         int64_t n_nbr) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n_nbr;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    // runs in a separate launch after the claim kernel, so every table
+    // entry is a published id by now
+    out[i] = table[nbrs[i]];
+  }
+}
+
+__global__ void compact_reset_kernel(int64_t* __restrict__ table,
+                                     const int64_t* __restrict__ srcdata,
+                                     int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    table[srcdata[i]] = -1;
+  }
+}
+
+std::tuple<at::Tensor, at::Tensor> compact_ids(at::Tensor table,
+                                               at::Tensor seeds,
+                                               at::Tensor neighbors) {
+  TORCH_CHECK(table.is_cuda(), "compact_ids: GPU tensors expected");
+  const int64_t n_seed = seeds.numel();
+  const int64_t n_nbr = neighbors.numel();
+  // worst case: every neighbor is new
+  auto srcdata = at::empty({n_seed + n_nbr}, seeds.options());
+  auto counter = at::zeros({1}, seeds.options());
+  auto local = at::empty({n_nbr}, seeds.options());
+  const int block = 256;
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(compact_seed_kernel, dim3(grid_for(n_seed, block)),
+                     dim3(block), 0, stream, table.data_ptr<int64_t>(),
+                     seeds.data_ptr<int64_t>(), srcdata.data_ptr<int64_t>(),
+                     n_seed);
+  if (n_nbr > 0) {
+    hipLaunchKernelGGL(compact_claim_kernel, dim3(grid_for(n_nbr, block)),
+                       dim3(block), 0, stream, table.data_ptr<int64_t>(),
+                       neighbors.data_ptr<int64_t>(),
+                       srcdata.data_ptr<int64_t>(),
+                       reinterpret_cast<unsigned long long*>(
+                           counter.data_ptr<int64_t>()),
+                       n_seed, n_nbr);
+    hipLaunchKernelGGL(compact_translate_kernel,
+                       dim3(grid_for(n_nbr, block)), dim3(block), 0, stream,
+                       table.data_ptr<int64_t>(),
+                       neighbors.data_ptr<int64_t>(),
+                       local.data_ptr<int64_t>(), n_nbr);
+  }
+  DOA_CHECK_HIP(hipGetLastError());
+  const int64_t n_new = counter.item<int64_t>();  // syncs the stream
+  srcdata = srcdata.narrow(0, 0, n_seed + n_new);
+  hipLaunchKernelGGL(compact_reset_kernel,
+                     dim3(grid_for(n_seed + n_new, block)), dim3(block), 0,
+                     stream, table.data_ptr<int64_t>(),
+                     srcdata.data_ptr<int64_t>(), n_seed + n_new);
+  DOA_CHECK_HIP(hipGetLastError());
+  return std::make_tuple(srcdata, local);
+}
+
+}  // namespace doa

@@ -75,6 +75,9 @@ class CompactionWorkspace:
     ) -> Tuple[torch.Tensor, torch.Tensor]:
         """Returns (srcdata_nids, local_neighbor_ids); seeds occupy local ids
         [0, len(seeds)) — the DGL block convention (dst nodes first)."""
+        if seeds.is_cuda:
+            ext = backend.ext_for(seeds)
+            return ext.compact_ids(self.table, seeds, neighbors)
         n_seed = seeds.numel()
         t = self.table
         t[seeds] = torch.arange(n_seed, device=seeds.device)

@@ -197,3 +197,29 @@ def test_graphsage_end_to_end_gpu(dev):
         model_gpu.named_parameters(), model_cpu.named_parameters()
     ):
         assert torch.allclose(p1.grad.cpu(), p2.grad, atol=1e-3, rtol=1e-3), n1
+
+
+def test_compact_ids_gpu(dev):
+    from dgl_operator_amd.ops.sampling import CompactionWorkspace
+
+    N = 10_000
+    ws = CompactionWorkspace(N, dev)
+    seeds = torch.randperm(N, device=dev)[:500]
+    nbrs = torch.randint(0, N, (20_000,), device=dev)
+    srcdata, local = ws.relabel(seeds, nbrs)
+    # seeds first, in order
+    assert torch.equal(srcdata[:500], seeds)
+    # translation is consistent: srcdata[local[i]] == nbrs[i]
+    assert torch.equal(srcdata[local], nbrs)
+    # srcdata ids are unique
+    assert torch.unique(srcdata).numel() == srcdata.numel()
+    # set of new nodes matches CPU reference
+    ref_new = torch.unique(nbrs.cpu()[~torch.isin(nbrs.cpu(), seeds.cpu())])
+    got_new = srcdata[500:].cpu().sort().values
+    assert torch.equal(got_new, ref_new)
+    # table fully reset
+    assert int((ws.table != -1).sum()) == 0
+    # reusable: second call works
+    srcdata2, local2 = ws.relabel(seeds, nbrs)
+    assert srcdata2.numel() == srcdata.numel()
+    assert torch.equal(srcdata2[local2], nbrs)
