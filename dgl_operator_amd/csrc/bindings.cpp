@@ -25,6 +25,17 @@ void sparse_adagrad(at::Tensor emb, at::Tensor state, at::Tensor ids,
 at::Tensor ldg_partition(at::Tensor indptr, at::Tensor indices,
                          at::Tensor cindptr, at::Tensor cindices,
                          int64_t num_parts);
+at::Tensor pdist_neg_fwd(at::Tensor base, at::Tensor neg, int64_t p,
+                         double gamma);
+std::tuple<at::Tensor, at::Tensor> pdist_neg_bwd(at::Tensor base,
+                                                 at::Tensor neg,
+                                                 at::Tensor out,
+                                                 at::Tensor gout, int64_t p,
+                                                 double gamma);
+at::Tensor cpdist_neg_fwd(at::Tensor base_r, at::Tensor base_i, at::Tensor neg,
+                          double gamma);
+std::tuple<at::Tensor, at::Tensor, at::Tensor> cpdist_neg_bwd(
+    at::Tensor base_r, at::Tensor base_i, at::Tensor neg, at::Tensor gout);
 
 }  // namespace doa
 
@@ -41,4 +52,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("compact_ids", &doa::compact_ids);
   m.def("sparse_adagrad", &doa::sparse_adagrad);
   m.def("ldg_partition", &doa::ldg_partition);
+  m.def("pdist_neg_fwd", &doa::pdist_neg_fwd);
+  m.def("pdist_neg_bwd", &doa::pdist_neg_bwd);
+  m.def("cpdist_neg_fwd", &doa::cpdist_neg_fwd);
+  m.def("cpdist_neg_bwd", &doa::cpdist_neg_bwd);
 }

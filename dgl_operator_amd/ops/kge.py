@@ -23,7 +23,54 @@ from typing import Optional
 import torch
 import torch.nn.functional as F
 
+from . import backend
+
 SCORE_FUNCS = {}
+
+
+class _PDistNeg(torch.autograd.Function):
+    """Fused pairwise-distance neg score (HIP kernel, csrc/kge.hip):
+    out[C,c,n] = gamma - ||base[C,c,:] - neg[C,n,:]||_p without materializing
+    the [C,c,n,D] broadcast."""
+
+    @staticmethod
+    def forward(ctx, base, neg, p, gamma):
+        ext = backend.ext_for(base)
+        base, neg = base.contiguous(), neg.contiguous()
+        out = ext.pdist_neg_fwd(base, neg, p, gamma)
+        ctx.save_for_backward(base, neg, out)
+        ctx.p, ctx.gamma = p, gamma
+        return out
+
+    @staticmethod
+    def backward(ctx, gout):
+        base, neg, out = ctx.saved_tensors
+        ext = backend.ext_for(base)
+        gb, gn = ext.pdist_neg_bwd(base, neg, out, gout.contiguous(), ctx.p,
+                                   ctx.gamma)
+        return gb, gn, None, None
+
+
+class _CPDistNeg(torch.autograd.Function):
+    """Fused complex-modulus pairwise distance (RotatE)."""
+
+    @staticmethod
+    def forward(ctx, base_r, base_i, neg, gamma):
+        ext = backend.ext_for(base_r)
+        base_r, base_i, neg = (
+            base_r.contiguous(), base_i.contiguous(), neg.contiguous()
+        )
+        out = ext.cpdist_neg_fwd(base_r, base_i, neg, gamma)
+        ctx.save_for_backward(base_r, base_i, neg)
+        return out
+
+    @staticmethod
+    def backward(ctx, gout):
+        base_r, base_i, neg = ctx.saved_tensors
+        ext = backend.ext_for(base_r)
+        gbr, gbi, gn = ext.cpdist_neg_bwd(base_r, base_i, neg,
+                                          gout.contiguous())
+        return gbr, gbi, gn, None
 
 
 def _register(name):
@@ -66,6 +113,7 @@ class ScoreFunc:
 class TransEL2(ScoreFunc):
     name = "TransE_l2"
     gamma_default = 12.0
+    p = 2
 
     def __init__(self, gamma: float = 12.0):
         self.gamma = gamma
@@ -75,19 +123,24 @@ class TransEL2(ScoreFunc):
 
     def _bcast(self, h, r, t, neg_head):
         d = (h + r - t) if not neg_head else (t + r - h)
-        return self.gamma - torch.norm(d, p=2, dim=-1)
+        return self.gamma - torch.norm(d, p=self.p, dim=-1)
+
+    def neg(self, head, rel, neg_tail, neg_head: bool = False):
+        # ||h + r - t_neg|| (neg tail) and ||h_neg - (t - r)|| (neg head,
+        # `head` arg carries the tail rows) share one pairwise-distance form
+        if head.is_cuda and backend.has_extension():
+            base = head - rel if neg_head else head + rel
+            return _PDistNeg.apply(base, neg_tail, self.p, self.gamma)
+        return super().neg(head, rel, neg_tail, neg_head)
 
 
 @_register("TransE_l1")
 class TransEL1(TransEL2):
     name = "TransE_l1"
+    p = 1
 
     def edge(self, head, rel, tail):
         return self.gamma - torch.norm(head + rel - tail, p=1, dim=-1)
-
-    def _bcast(self, h, r, t, neg_head):
-        d = (h + r - t) if not neg_head else (t + r - h)
-        return self.gamma - torch.norm(d, p=1, dim=-1)
 
 
 @_register("DistMult")
@@ -169,6 +222,21 @@ class RotatE(ScoreFunc):
         return self.gamma - torch.sqrt(dr * dr + di * di).sum(-1)
 
     def neg(self, head, rel, neg_tail, neg_head: bool = False):
+        # one fused pairwise form for both corruption sides: rotation is an
+        # isometry, so ||rot(neg) - tail|| = ||neg - rot^-1(tail)|| — corrupt
+        # head uses the tail rotated by MINUS the phase as the base
+        if head.is_cuda and backend.has_extension():
+            D2 = head.shape[-1] // 2
+            ph = self._phase(rel)
+            rr, ri = torch.cos(ph), torch.sin(ph)
+            hr, hi = head[..., :D2], head[..., D2:]
+            if neg_head:  # `head` carries tail rows; rotate by -phase
+                base_r = hr * rr + hi * ri
+                base_i = hi * rr - hr * ri
+            else:
+                base_r = hr * rr - hi * ri
+                base_i = hr * ri + hi * rr
+            return _CPDistNeg.apply(base_r, base_i, neg_tail, self.gamma)
         h = head.unsqueeze(2)
         r = rel.unsqueeze(2)
         t = neg_tail.unsqueeze(1)

@@ -223,3 +223,35 @@ def test_compact_ids_gpu(dev):
     srcdata2, local2 = ws.relabel(seeds, nbrs)
     assert srcdata2.numel() == srcdata.numel()
     assert torch.equal(srcdata2[local2], nbrs)
+
+
+@pytest.mark.parametrize("name", ["TransE_l1", "TransE_l2", "RotatE"])
+@pytest.mark.parametrize("neg_head", [False, True])
+def test_kge_fused_neg_matches_cpu(dev, name, neg_head):
+    """Fused HIP pairwise neg-score (fwd+bwd) vs the CPU broadcast reference."""
+    from dgl_operator_amd.ops import get_score_func
+
+    C, c, n, D = 3, 8, 16, 32
+    Dr = D // 2 if name == "RotatE" else D
+    fn = get_score_func(name, gamma=10.0)
+    head = torch.randn(C, c, D)
+    rel = torch.randn(C, c, Dr)
+    negs = torch.randn(C, n, D)
+
+    h_cpu = head.clone().requires_grad_(True)
+    r_cpu = rel.clone().requires_grad_(True)
+    n_cpu = negs.clone().requires_grad_(True)
+    out_cpu = fn.neg(h_cpu, r_cpu, n_cpu, neg_head=neg_head)
+    gout = torch.randn_like(out_cpu)
+    out_cpu.backward(gout)
+
+    h_gpu = head.to(dev).requires_grad_(True)
+    r_gpu = rel.to(dev).requires_grad_(True)
+    n_gpu = negs.to(dev).requires_grad_(True)
+    out_gpu = fn.neg(h_gpu, r_gpu, n_gpu, neg_head=neg_head)
+    out_gpu.backward(gout.to(dev))
+
+    assert torch.allclose(out_gpu.cpu(), out_cpu, atol=1e-4, rtol=1e-4)
+    assert torch.allclose(h_gpu.grad.cpu(), h_cpu.grad, atol=1e-3, rtol=1e-3)
+    assert torch.allclose(r_gpu.grad.cpu(), r_cpu.grad, atol=1e-3, rtol=1e-3)
+    assert torch.allclose(n_gpu.grad.cpu(), n_cpu.grad, atol=1e-3, rtol=1e-3)
