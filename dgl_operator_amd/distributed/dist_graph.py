@@ -196,5 +196,23 @@ class DistGraph:
         )
         feat.index_add_(0, dest_ids - self.lo, dest_rows)
 
+    def full_neighbor_block(self, seeds: torch.Tensor) -> Block:
+        """Block over ALL in-edges of the (owned) seeds — the layer-wise
+        full-neighbor inference structure (DistSAGE.inference parity)."""
+        local = seeds - self.lo
+        starts = self.csc_indptr[local]
+        counts = self.csc_indptr[local + 1] - starts
+        total = int(counts.sum())
+        if total:
+            off = _cumsum0(counts)
+            pos = torch.repeat_interleave(starts, counts) + (
+                torch.arange(total, device=seeds.device)
+                - torch.repeat_interleave(off[:-1], counts)
+            )
+            nbrs = self.csc_indices[pos]
+        else:
+            nbrs = self.csc_indices.new_empty(0)
+        return to_block(seeds, nbrs, counts, self.workspace)
+
     def barrier(self):
         comm.barrier()

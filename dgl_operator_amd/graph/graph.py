@@ -121,6 +121,60 @@ class Graph:
         g.ndata = dict(self.ndata)
         return g
 
+    # -- DGL-style message passing API ------------------------------------
+    def update_all(self, message_func, reduce_func) -> None:
+        """g.update_all(fn.copy_u('h','m'), fn.mean('m','h_N')) — builtin
+        pairs run the fused HIP gspmm; Python UDF pairs take the
+        degree-bucketing fallback (ops/udf.py)."""
+        _update_all(self, message_func, reduce_func, self._num_nodes)
+
+    def apply_edges(self, edge_func) -> None:
+        """g.apply_edges(fn.u_dot_v('h','h','score')) or a Python UDF over
+        EdgeBatch; writes into g.edata in ORIGINAL edge order."""
+        _apply_edges(self, edge_func)
+
+
+def _update_all(g, message_func, reduce_func, num_dst: int) -> None:
+    from .. import fn as fnmod
+    from ..ops import gspmm
+    from ..ops.udf import update_all_udf
+
+    if isinstance(message_func, fnmod.MessageFn) and isinstance(
+        reduce_func, fnmod.ReduceFn
+    ):
+        assert message_func.out_field == reduce_func.msg_field, (
+            "message out field must feed the reduce"
+        )
+        feat = g.ndata[message_func.src_field]
+        w = g.edata[message_func.edge_field] if message_func.op == "u_mul_e" else None
+        g.ndata[reduce_func.out_field] = gspmm(
+            g, message_func.op, reduce_func.op, feat, w
+        )
+        return
+    out = update_all_udf(g, g.ndata, g.edata, message_func, reduce_func, num_dst)
+    g.ndata.update(out)
+
+
+def _apply_edges(g, edge_func) -> None:
+    from .. import fn as fnmod
+    from ..ops import sddmm_dot
+    from ..ops.udf import EdgeBatch
+
+    if isinstance(edge_func, fnmod.EdgeFn):
+        assert edge_func.op == "u_dot_v"
+        g.edata[edge_func.out_field] = sddmm_dot(
+            g, g.ndata[edge_func.lhs_field], g.ndata[edge_func.rhs_field]
+        )
+        return
+    src, dst = g.edges()
+    batch = EdgeBatch(
+        {k: v[src] for k, v in g.ndata.items()},
+        {k: v[dst] for k, v in g.ndata.items()},
+        dict(g.edata),
+    )
+    for k, v in edge_func(batch).items():
+        g.edata[k] = v
+
 
 class Block:
     """Bipartite message-flow graph (MFG) produced by neighbor sampling.

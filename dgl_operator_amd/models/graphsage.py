@@ -48,3 +48,38 @@ class GraphSAGE(nn.Module):
                 h = F.relu(h)
                 h = self.dropout(h)
         return h
+
+
+@torch.no_grad()
+def inference_dist(model: GraphSAGE, dg, batch_size: int = 1000,
+                   feat_key: str = "feat") -> torch.Tensor:
+    """Layer-wise full-neighbor distributed inference — parity with
+    DistSAGE.inference (/root/reference/examples/GraphSAGE_dist/code/
+    train_dist.py:96-144): per layer, each rank computes the layer output for
+    its OWNED nodes in seed batches over full in-neighbor blocks, pulling the
+    previous layer's output from peer shards (the reference's DistTensor
+    writes + g.barrier()); a barrier separates layers. Returns this rank's
+    shard of the final layer output."""
+    model.eval()
+    cur_key = feat_key
+    owned = dg.owned_nodes()
+    n_layers = len(model.layers)
+    for li, layer in enumerate(model.layers):
+        outs = []
+        for s in range(0, owned.numel(), batch_size):
+            seeds = owned[s : s + batch_size]
+            blk = dg.full_neighbor_block(seeds)
+            x = dg.pull(cur_key, blk.srcdata_nids)
+            h = layer(blk, x)
+            if li != n_layers - 1:
+                h = F.relu(h)
+            outs.append(h)
+        next_key = f"__infer_h{li}"
+        dg.ndata[next_key] = torch.cat(outs) if outs else None
+        dg.barrier()
+        cur_key = next_key
+    out = dg.ndata[cur_key]
+    # clean intermediate layer shards except the final
+    for li in range(n_layers - 1):
+        dg.ndata.pop(f"__infer_h{li}", None)
+    return out

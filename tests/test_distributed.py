@@ -175,3 +175,26 @@ def _train_worker(rank, world):
 
 def test_dist_train_step():
     _run_workers(_train_worker)
+
+
+def _inference_worker(rank, world):
+    import torch.nn.functional as F
+
+    from dgl_operator_amd.models import GraphSAGE
+    from dgl_operator_amd.models.graphsage import inference_dist
+
+    g, dg = _make_shard(rank, world)
+    torch.manual_seed(7)
+    model = GraphSAGE(8, 16, 4, n_layers=2, dropout=0.0)
+    for p in model.parameters():
+        dist.broadcast(p.data, src=0)
+    shard = inference_dist(model, dg, batch_size=37)
+    # reference: full-graph forward on the whole graph (eval mode)
+    model.eval()
+    with torch.no_grad():
+        full = model(g, g.ndata["feat"])
+    assert torch.allclose(shard, full[dg.lo : dg.hi], atol=1e-4, rtol=1e-4)
+
+
+def test_dist_layerwise_inference():
+    _run_workers(_inference_worker)
