@@ -132,3 +132,62 @@ class KGEdgeSampler:
         neg_head = self.step % 2 == 1
         self.step += 1
         return self.h[idx], self.r[idx], self.t[idx], negs, neg_head
+
+
+@torch.no_grad()
+def evaluate_kge(
+    model: DistKGEModel,
+    heads: torch.Tensor,
+    rels: torch.Tensor,
+    tails: torch.Tensor,
+    batch_size: int = 128,
+    hits: Tuple[int, ...] = (1, 3, 10),
+    corrupt: str = "tail",
+):
+    """Raw-setting link-prediction metrics (MRR, MR, Hits@K) — the dglke eval
+    protocol (reference EvalSampler/EvalDataset, hotfix/sampler.py:514-821):
+    rank the true entity among ALL entities. With sharded entity tables each
+    rank scores the candidates it OWNS and the global rank is the all-reduced
+    count of higher-scoring candidates + 1.
+    """
+    import torch.distributed as dist
+
+    device = heads.device
+    rank_sum = 0.0
+    rr_sum = 0.0
+    hit_counts = {k: 0.0 for k in hits}
+    n_total = heads.numel()
+    shard = model.entities.local  # [M, D] owned candidate entities
+    M = shard.shape[0]
+    for s in range(0, n_total, batch_size):
+        hh = heads[s : s + batch_size]
+        rr = rels[s : s + batch_size]
+        tt = tails[s : s + batch_size]
+        B = hh.numel()
+        h = model.entities.pull(hh)
+        r = model.relations.pull(rr)
+        t = model.entities.pull(tt)
+        true_score = model.score.edge(h, r, t)  # [B]
+        if corrupt == "tail":
+            cand = model.score.neg(
+                h.view(1, B, -1), r.view(1, B, -1), shard.view(1, M, -1),
+                neg_head=False,
+            )[0]  # [B, M]
+        else:
+            cand = model.score.neg(
+                t.view(1, B, -1), r.view(1, B, -1), shard.view(1, M, -1),
+                neg_head=True,
+            )[0]
+        higher = (cand > true_score.unsqueeze(1)).sum(1).double()
+        if dist.is_available() and dist.is_initialized():
+            dist.all_reduce(higher)
+        ranks = higher + 1.0
+        rank_sum += float(ranks.sum())
+        rr_sum += float((1.0 / ranks).sum())
+        for k in hits:
+            hit_counts[k] += float((ranks <= k).sum())
+    return {
+        "MR": rank_sum / n_total,
+        "MRR": rr_sum / n_total,
+        **{f"Hits@{k}": hit_counts[k] / n_total for k in hits},
+    }

@@ -28,6 +28,7 @@ class Pod:
     annotations: Dict[str, str] = field(default_factory=dict)
     spec: Dict[str, Any] = field(default_factory=dict)
     phase: PodPhase = PodPhase.PENDING
+    reason: Optional[str] = None  # e.g. "Evicted"
     ip: Optional[str] = None
     containers_ready: bool = False
     owner: Optional[str] = None  # owning DGLJob name
@@ -137,9 +138,11 @@ class FakeCluster(Cluster):
 
     # -- fake kubelet ------------------------------------------------------
     def set_pod_phase(self, namespace, name, phase: PodPhase,
-                      containers_ready: Optional[bool] = None):
+                      containers_ready: Optional[bool] = None,
+                      reason: Optional[str] = None):
         p = self.pods[self._key(namespace, name)]
         p.phase = phase
+        p.reason = reason
         if phase == PodPhase.RUNNING and p.ip is None:
             p.ip = f"10.244.0.{self._next_ip}"
             self._next_ip += 1

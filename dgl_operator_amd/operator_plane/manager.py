@@ -1,0 +1,145 @@
+"""Operator manager process — the main.go equivalent
+(/root/reference/main.go:52-105): flag surface (--metrics-bind-address,
+--health-probe-bind-address, --reconcile-interval), Prometheus metrics,
+healthz/readyz endpoints, and the reconcile loop over submitted DGLJobs.
+
+controller-runtime's informer machinery is k8s-specific; the observable
+contract — every job reconciled on change/interval, metrics exported,
+health probes — is implemented against the Cluster abstraction so the same
+manager drives the FakeCluster in tests and a kubectl-backed cluster in a
+real deployment.
+"""
+from __future__ import annotations
+
+import argparse
+import threading
+import time
+from http.server import BaseHTTPRequestHandler, HTTPServer
+from typing import Dict, Optional
+
+from .api import DGLJob, JobPhase, job_from_manifest
+from .cluster import Cluster, FakeCluster
+from .reconciler import DGLJobReconciler
+
+try:
+    from prometheus_client import Counter, Gauge, Histogram, start_http_server
+
+    _PROM = True
+    RECONCILE_TOTAL = Counter(
+        "dgljob_reconcile_total", "Total DGLJob reconciles", ["job"]
+    )
+    RECONCILE_ERRORS = Counter(
+        "dgljob_reconcile_errors_total", "Failed reconciles", ["job"]
+    )
+    RECONCILE_SECONDS = Histogram(
+        "dgljob_reconcile_duration_seconds", "Reconcile wall time"
+    )
+    JOB_PHASE = Gauge("dgljob_phase", "Job phase (1=active)", ["job", "phase"])
+except ImportError:  # pragma: no cover
+    _PROM = False
+
+
+class _HealthHandler(BaseHTTPRequestHandler):
+    manager: "Manager" = None
+
+    def do_GET(self):  # noqa: N802
+        ok = self.path in ("/healthz", "/readyz") and (
+            self.path == "/healthz" or self.manager.ready
+        )
+        self.send_response(200 if ok else 503)
+        self.end_headers()
+        self.wfile.write(b"ok" if ok else b"not ready")
+
+    def log_message(self, *a):  # silence
+        pass
+
+
+class Manager:
+    def __init__(self, cluster: Optional[Cluster] = None,
+                 reconcile_interval: float = 0.5):
+        self.cluster = cluster or FakeCluster()
+        self.reconciler = DGLJobReconciler(self.cluster)
+        self.jobs: Dict[str, DGLJob] = {}
+        self.interval = reconcile_interval
+        self.ready = False
+        self._stop = threading.Event()
+
+    # -- job API (the CRD surface) -----------------------------------------
+    def submit(self, manifest) -> DGLJob:
+        job = manifest if isinstance(manifest, DGLJob) else job_from_manifest(manifest)
+        self.jobs[f"{job.namespace}/{job.name}"] = job
+        return job
+
+    def delete(self, namespace: str, name: str):
+        job = self.jobs.get(f"{namespace}/{name}")
+        if job:
+            job.deletion_timestamp = time.time()
+
+    def get(self, namespace: str, name: str) -> Optional[DGLJob]:
+        return self.jobs.get(f"{namespace}/{name}")
+
+    # -- reconcile loop ----------------------------------------------------
+    def reconcile_once(self):
+        for key, job in list(self.jobs.items()):
+            t0 = time.time()
+            try:
+                self.reconciler.reconcile(job)
+                if _PROM:
+                    RECONCILE_TOTAL.labels(job=job.name).inc()
+                    RECONCILE_SECONDS.observe(time.time() - t0)
+                    for ph in JobPhase:
+                        JOB_PHASE.labels(job=job.name, phase=ph.value).set(
+                            1.0 if job.status.phase == ph else 0.0
+                        )
+            except Exception:  # noqa: BLE001
+                if _PROM:
+                    RECONCILE_ERRORS.labels(job=job.name).inc()
+                raise
+            if job.deletion_timestamp is not None:
+                del self.jobs[key]
+
+    def run(self, metrics_port: Optional[int] = None,
+            health_port: Optional[int] = None, block: bool = True):
+        if metrics_port and _PROM:
+            start_http_server(metrics_port)
+        health_server = None
+        if health_port:
+            _HealthHandler.manager = self
+            health_server = HTTPServer(("127.0.0.1", health_port), _HealthHandler)
+            threading.Thread(target=health_server.serve_forever,
+                             daemon=True).start()
+        self.ready = True
+        if not block:
+            threading.Thread(target=self._loop, daemon=True).start()
+            return
+        self._loop()
+
+    def _loop(self):
+        while not self._stop.is_set():
+            self.reconcile_once()
+            self._stop.wait(self.interval)
+
+    def stop(self):
+        self._stop.set()
+
+
+def main(argv=None):
+    p = argparse.ArgumentParser(prog="dgl-operator-manager")
+    p.add_argument("--metrics-bind-address", default=":8080")
+    p.add_argument("--health-probe-bind-address", default=":8081")
+    p.add_argument("--reconcile-interval", type=float, default=0.5)
+    p.add_argument("--job", action="append", default=[],
+                   help="DGLJob manifest YAML file(s) to manage")
+    args = p.parse_args(argv)
+    mgr = Manager(reconcile_interval=args.reconcile_interval)
+    for path in args.job:
+        with open(path) as f:
+            mgr.submit(f.read())
+    mgr.run(
+        metrics_port=int(args.metrics_bind_address.rsplit(":", 1)[1]),
+        health_port=int(args.health_probe_bind_address.rsplit(":", 1)[1]),
+    )
+
+
+if __name__ == "__main__":
+    main()

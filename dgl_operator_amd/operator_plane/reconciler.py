@@ -71,6 +71,17 @@ class DGLJobReconciler:
             self._cleanup(job, job.spec.clean_pod_policy)
             return job
 
+        # -- eviction retry (dgljob_controller.go:146-172): an EVICTED
+        # launcher of an incomplete job is deleted and recreated below,
+        # instead of failing the job ------------------------------------
+        launcher = c.get_pod(job.namespace, job.launcher_name())
+        if (
+            launcher is not None
+            and launcher.phase == PodPhase.FAILED
+            and launcher.reason == "Evicted"
+        ):
+            c.delete_pod(job.namespace, launcher.name)
+
         if job.status.start_time is None:
             job.status.start_time = time.time()
 

@@ -103,3 +103,36 @@ def test_checkpoint_roundtrip(tmp_path):
     emb2.load_shard(p)
     assert torch.allclose(emb.local, emb2.local)
     assert torch.allclose(emb.state, emb2.state)
+
+
+def _eval_worker(rank, world):
+    from dgl_operator_amd.distributed import DistKGEModel, KGEdgeSampler
+    from dgl_operator_amd.distributed.kge import evaluate_kge
+
+    torch.manual_seed(0)
+    E, R, D = 120, 4, 16
+    h = torch.randint(0, E, (600,))
+    r = torch.randint(0, R, (600,))
+    t = torch.randint(0, E, (600,))
+    model = DistKGEModel(E, R, D, score_func="TransE_l2", gamma=8.0,
+                         rank=rank, world_size=world)
+    sampler = KGEdgeSampler((h, r, t), E, batch_size=64, neg_sample_size=16,
+                            chunk_size=16, seed=rank)
+    m0 = evaluate_kge(model, h[:100], r[:100], t[:100])
+    for step in range(60):
+        hh, rr, tt, negs, neg_head = sampler.next_batch()
+        model.train_step(hh, rr, tt, negs, chunk_size=16, lr=0.1,
+                         neg_head=neg_head)
+    m1 = evaluate_kge(model, h[:100], r[:100], t[:100])
+    assert 0.0 < m0["MRR"] <= 1.0
+    assert m1["MRR"] > m0["MRR"]  # training improves ranking of true triples
+    assert m1["Hits@10"] >= m0["Hits@10"] - 0.05
+    # identical metrics on every rank (they are global reductions)
+    v = torch.tensor([m1["MRR"]])
+    ref = v.clone()
+    dist.broadcast(ref, src=0)
+    assert torch.allclose(v, ref, atol=1e-9)
+
+
+def test_kge_eval_metrics():
+    _run_workers(_eval_worker)

@@ -242,3 +242,54 @@ def test_watcher_ready_and_finished():
     assert watcher.watch(c, "default", names, "finished", timeout=2)
     # launcher entries are skipped
     assert watcher.parse_watchfile(cm.data["leadfile"]) == []
+
+
+def test_evicted_launcher_is_retried():
+    c = FakeCluster()
+    r = DGLJobReconciler(c)
+    job = make_job()
+    r.reconcile(job)
+    c.set_pod_phase("default", "graphsage-dist-launcher", PodPhase.FAILED,
+                    reason="Evicted")
+    r.reconcile(job)
+    # launcher recreated instead of failing the job
+    lp = c.get_pod("default", "graphsage-dist-launcher")
+    assert lp is not None and lp.phase == PodPhase.PENDING
+    assert job.status.phase != JobPhase.FAILED
+
+
+def test_manager_loop_and_health(tmp_path):
+    import urllib.request
+
+    from dgl_operator_amd.operator_plane.manager import Manager
+
+    c = FakeCluster()
+    mgr = Manager(cluster=c, reconcile_interval=0.05)
+    job = mgr.submit(GRAPHSAGE_YAML)
+    port = 18125
+    mgr.run(health_port=port, block=False)
+    try:
+        import time
+
+        time.sleep(0.3)
+        assert c.get_pod("default", "graphsage-dist-launcher") is not None
+        # drive to completion through the fake kubelet
+        c.set_pod_phase("default", "graphsage-dist-partitioner",
+                        PodPhase.SUCCEEDED)
+        time.sleep(0.2)
+        c.run_all_pending()
+        time.sleep(0.2)
+        c.set_pod_phase("default", "graphsage-dist-launcher",
+                        PodPhase.SUCCEEDED)
+        time.sleep(0.2)
+        assert mgr.get("default", "graphsage-dist").status.phase == JobPhase.COMPLETED
+        with urllib.request.urlopen(f"http://127.0.0.1:{port}/healthz") as resp:
+            assert resp.status == 200
+        with urllib.request.urlopen(f"http://127.0.0.1:{port}/readyz") as resp:
+            assert resp.status == 200
+        # deletion: pods cleaned, job dropped from the store
+        mgr.delete("default", "graphsage-dist")
+        time.sleep(0.3)
+        assert mgr.get("default", "graphsage-dist") is None
+    finally:
+        mgr.stop()
