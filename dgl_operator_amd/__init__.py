@@ -31,4 +31,7 @@ tools         dglrun workflow CLI, launch/dispatch/revise_hostfile
 
 __version__ = "0.1.0"
 
-from . import graph  # noqa: F401
+try:  # compute plane needs torch; the control-plane-only manager image
+    from . import graph  # noqa: F401
+except ImportError:  # pragma: no cover
+    graph = None

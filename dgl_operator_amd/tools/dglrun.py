@@ -52,12 +52,25 @@ def build_parser():
     p.add_argument("--hostfile", default="/etc/dgl/hostfile")
     p.add_argument("--leadfile", default="/etc/dgl/leadfile")
     p.add_argument("--master-port", type=int, default=29400)
+    p.add_argument("--ignore-partition", action="store_true",
+                   help="reuse an existing partition under workspace/dataset "
+                        "(the reference dglkerun's PVC-reuse path)")
     return p
 
 
 def run_partitioner(args):
     dataset = os.path.join(args.workspace, "dataset")
     os.makedirs(dataset, exist_ok=True)
+    if args.ignore_partition and os.path.exists(
+        os.path.join(dataset, f"{args.graph_name}.json")
+    ):
+        print("[dglrun] Phase 1/5 skipped (--ignore-partition)", flush=True)
+    else:
+        run_phase1(args, dataset)
+    _deliver(args, dataset)
+
+
+def run_phase1(args, dataset):
     with phase("Phase 1/5 partition"):
         _run(
             f"python {args.partition_entry_point} "
@@ -65,6 +78,9 @@ def run_partitioner(args):
             f"--num-partitions {args.num_partitions} "
             f"--output {dataset} {args.partition_entry_args}"
         )
+
+
+def _deliver(args, dataset):
     with phase("Phase 2/5 deliver"):
         with open(args.leadfile) as f:
             leads = parse_hostfile(f.read())

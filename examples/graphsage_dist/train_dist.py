@@ -41,6 +41,8 @@ def parse_args():
     p.add_argument("--dropout", type=float, default=0.5)
     p.add_argument("--log-every", type=int, default=20)
     p.add_argument("--eval-every", type=int, default=0)
+    p.add_argument("--checkpoint-path", type=str, default="",
+                   help="save model+optimizer per epoch; resume if present")
     return p.parse_args()
 
 
@@ -79,6 +81,22 @@ def main():
     opt = torch.optim.Adam(model.parameters(), lr=args.lr)
     fanouts = [int(x) for x in args.fan_out.split(",")]
 
+    # checkpoint/resume (the operator's de-facto resumable artifact is the
+    # partition dir; model state adds true training resume on top)
+    start_epoch = 0
+    ckpt_file = None
+    if args.checkpoint_path:
+        os.makedirs(args.checkpoint_path, exist_ok=True)
+        ckpt_file = os.path.join(args.checkpoint_path, "graphsage.pt")
+        if os.path.exists(ckpt_file):
+            state = torch.load(ckpt_file, map_location=device,
+                               weights_only=True)
+            model.load_state_dict(state["model"])
+            opt.load_state_dict(state["optimizer"])
+            start_epoch = state["epoch"] + 1
+            if rank == 0:
+                print(f"resumed from epoch {state['epoch']}", flush=True)
+
     train_mask = dg.ndata.get("train_mask")
     owned = dg.owned_nodes()
     train_nids = owned[train_mask.bool()] if train_mask is not None else owned
@@ -87,7 +105,7 @@ def main():
     gen.manual_seed(1234 + rank)
     steps_per_epoch = max(1, train_nids.numel() // args.batch_size)
 
-    for epoch in range(args.num_epochs):
+    for epoch in range(start_epoch, args.num_epochs):
         t_epoch = time.time()
         for step in range(steps_per_epoch):
             tic = time.time()
@@ -123,6 +141,12 @@ def main():
         if rank == 0:
             print(f"Epoch {epoch:03d} time {time.time() - t_epoch:.2f}s",
                   flush=True)
+            if ckpt_file:
+                torch.save(
+                    {"model": model.state_dict(),
+                     "optimizer": opt.state_dict(), "epoch": epoch},
+                    ckpt_file,
+                )
 
     if dist.is_initialized():
         dist.destroy_process_group()
