@@ -206,3 +206,149 @@ std::tuple<at::Tensor, at::Tensor> compact_ids(at::Tensor table,
 }
 
 }  // namespace doa
+
+namespace doa {
+
+// ---------------------------------------------------------------------------
+// Fused sample+compact block builder — ONE host sync per hop.
+//
+// The separate sample_neighbors -> masked_select -> compact_ids chain costs
+// two device syncs (masked_select's output sizing + the compaction counter
+// read) plus ~4 extra kernels per hop; in the minibatch regime the step is
+// launch/latency-bound, so this path samples into a padded [n, fanout]
+// buffer, compacts/translates in place, and returns device-side totals the
+// caller fetches with a single .cpu() copy.
+// ---------------------------------------------------------------------------
+
+__global__ void compact_claim_padded_kernel(
+    int64_t* __restrict__ table, const int64_t* __restrict__ padded,
+    const int64_t* __restrict__ counts, int64_t* __restrict__ srcdata,
+    unsigned long long* __restrict__ counter, int64_t n_seed, int64_t n,
+    int fanout) {
+  const int64_t total = n * fanout;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t row = i / fanout;
+    if ((i % fanout) >= counts[row]) continue;
+    const int64_t v = padded[i];
+    const long long old = atomicCAS(
+        reinterpret_cast<unsigned long long*>(&table[v]),
+        (unsigned long long)(-1LL), (unsigned long long)(-2LL));
+    if ((long long)old == -1LL) {
+      const int64_t id = n_seed + (int64_t)atomicAdd(counter, 1ull);
+      srcdata[id] = v;
+      atomicExch(reinterpret_cast<unsigned long long*>(&table[v]),
+                 (unsigned long long)id);
+    }
+  }
+}
+
+__global__ void translate_padded_kernel(const int64_t* __restrict__ table,
+                                        int64_t* __restrict__ padded,
+                                        const int64_t* __restrict__ counts,
+                                        int64_t n, int fanout) {
+  const int64_t total = n * fanout;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t row = i / fanout;
+    if ((i % fanout) >= counts[row]) continue;
+    padded[i] = table[padded[i]];
+  }
+}
+
+__global__ void reset_from_srcdata_kernel(int64_t* __restrict__ table,
+                                          const int64_t* __restrict__ srcdata,
+                                          const unsigned long long* counter,
+                                          int64_t n_seed) {
+  const int64_t n = n_seed + (int64_t)*counter;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    table[srcdata[i]] = -1;
+  }
+}
+
+__global__ void pack_padded_kernel(const int64_t* __restrict__ padded,
+                                   const int64_t* __restrict__ counts,
+                                   const int64_t* __restrict__ offsets,
+                                   int64_t* __restrict__ out, int64_t n,
+                                   int fanout) {
+  const int64_t total = n * fanout;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t row = i / fanout;
+    const int64_t j = i % fanout;
+    if (j < counts[row]) out[offsets[row] + j] = padded[i];
+  }
+}
+
+// returns (padded_local [n,fanout], counts [n], srcdata [n_seed+n*fanout],
+//          totals [1] = n_new)  — all device tensors, no host sync here.
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> sample_block(
+    at::Tensor indptr, at::Tensor indices, at::Tensor table, at::Tensor seeds,
+    int64_t fanout, bool replace, int64_t seed) {
+  TORCH_CHECK(seeds.is_cuda(), "sample_block: GPU tensors expected");
+  const int64_t n = seeds.numel();
+  auto padded = at::empty({n, fanout}, seeds.options());
+  auto counts = at::empty({n}, seeds.options());
+  auto srcdata = at::empty({n + n * fanout}, seeds.options());
+  auto counter = at::zeros({1}, seeds.options());
+  const int block = 256;
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  if (replace) {
+    hipLaunchKernelGGL((sample_kernel<true>), dim3(grid_for(n, block)),
+                       dim3(block), 0, stream, indptr.data_ptr<int64_t>(),
+                       indices.data_ptr<int64_t>(), seeds.data_ptr<int64_t>(),
+                       padded.data_ptr<int64_t>(), counts.data_ptr<int64_t>(),
+                       n, (int)fanout, (uint64_t)seed);
+  } else {
+    hipLaunchKernelGGL((sample_kernel<false>), dim3(grid_for(n, block)),
+                       dim3(block), 0, stream, indptr.data_ptr<int64_t>(),
+                       indices.data_ptr<int64_t>(), seeds.data_ptr<int64_t>(),
+                       padded.data_ptr<int64_t>(), counts.data_ptr<int64_t>(),
+                       n, (int)fanout, (uint64_t)seed);
+  }
+  hipLaunchKernelGGL(compact_seed_kernel, dim3(grid_for(n, block)),
+                     dim3(block), 0, stream, table.data_ptr<int64_t>(),
+                     seeds.data_ptr<int64_t>(), srcdata.data_ptr<int64_t>(),
+                     n);
+  hipLaunchKernelGGL(compact_claim_padded_kernel,
+                     dim3(grid_for(n * fanout, block)), dim3(block), 0,
+                     stream, table.data_ptr<int64_t>(),
+                     padded.data_ptr<int64_t>(), counts.data_ptr<int64_t>(),
+                     srcdata.data_ptr<int64_t>(),
+                     reinterpret_cast<unsigned long long*>(
+                         counter.data_ptr<int64_t>()),
+                     n, n, (int)fanout);
+  hipLaunchKernelGGL(translate_padded_kernel,
+                     dim3(grid_for(n * fanout, block)), dim3(block), 0,
+                     stream, table.data_ptr<int64_t>(),
+                     padded.data_ptr<int64_t>(), counts.data_ptr<int64_t>(),
+                     n, (int)fanout);
+  hipLaunchKernelGGL(reset_from_srcdata_kernel,
+                     dim3(grid_for(n + n * fanout, block)), dim3(block), 0,
+                     stream, table.data_ptr<int64_t>(),
+                     srcdata.data_ptr<int64_t>(),
+                     reinterpret_cast<const unsigned long long*>(
+                         counter.data_ptr<int64_t>()),
+                     n);
+  DOA_CHECK_HIP(hipGetLastError());
+  return std::make_tuple(padded, counts, srcdata, counter);
+}
+
+at::Tensor pack_padded(at::Tensor padded, at::Tensor counts,
+                       at::Tensor offsets, int64_t total) {
+  const int64_t n = counts.numel();
+  const int fanout = padded.size(1);
+  auto out = at::empty({total}, padded.options());
+  const int block = 256;
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(pack_padded_kernel,
+                     dim3(grid_for(n * fanout, block)), dim3(block), 0,
+                     stream, padded.data_ptr<int64_t>(),
+                     counts.data_ptr<int64_t>(), offsets.data_ptr<int64_t>(),
+                     out.data_ptr<int64_t>(), n, fanout);
+  DOA_CHECK_HIP(hipGetLastError());
+  return out;
+}
+
+}  // namespace doa

@@ -21,7 +21,12 @@ import torch
 
 from ..graph.graph import Block, _coo_to_compressed
 from ..graph.partition import load_partition
-from ..ops.sampling import CompactionWorkspace, sample_neighbors, to_block
+from ..ops.sampling import (
+    CompactionWorkspace,
+    sample_block_fused,
+    sample_neighbors,
+    to_block,
+)
 from . import comm
 from .partition_book import PartitionBook
 
@@ -155,11 +160,23 @@ class DistGraph:
         owned by this rank). Returns (input_nodes, seeds, blocks)."""
         blocks: List[Block] = []
         cur = seeds
+        rank, ws = comm.world()
         for layer, fanout in enumerate(reversed(list(fanouts))):
-            nbrs, counts = self.sample_neighbors_dist(
-                cur, fanout, replace, seed=seed * 1000003 + layer
-            )
-            blk = to_block(cur, nbrs, counts, self.workspace)
+            if ws == 1 and cur.is_cuda:
+                # single-rank GPU fast path: fused sample+compact, local ids
+                # shifted into the global range
+                blk = sample_block_fused(
+                    self.csc_indptr, self.csc_indices, self.workspace,
+                    cur - self.lo if self.lo else cur, fanout, replace,
+                    seed=seed * 1000003 + layer,
+                )
+                if self.lo:
+                    blk.srcdata_nids = blk.srcdata_nids + self.lo
+            else:
+                nbrs, counts = self.sample_neighbors_dist(
+                    cur, fanout, replace, seed=seed * 1000003 + layer
+                )
+                blk = to_block(cur, nbrs, counts, self.workspace)
             blocks.insert(0, blk)
             cur = blk.srcdata_nids
         return cur, seeds, blocks

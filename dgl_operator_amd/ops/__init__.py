@@ -5,6 +5,7 @@ from .segment import segment_reduce, mean_nodes
 from .sampling import (
     sample_neighbors,
     to_block,
+    sample_block_fused,
     NeighborSampler,
     CompactionWorkspace,
 )
@@ -22,6 +23,7 @@ __all__ = [
     "mean_nodes",
     "sample_neighbors",
     "to_block",
+    "sample_block_fused",
     "NeighborSampler",
     "CompactionWorkspace",
     "get_score_func",

@@ -92,6 +92,40 @@ class CompactionWorkspace:
         return torch.cat([seeds, uniq_new]), local_nbrs
 
 
+def sample_block_fused(
+    indptr: torch.Tensor,
+    indices: torch.Tensor,
+    workspace: CompactionWorkspace,
+    seeds: torch.Tensor,
+    fanout: int,
+    replace: bool = False,
+    seed: int = 0,
+) -> Block:
+    """GPU fast path: fused sample+compact with ONE host sync per hop
+    (csrc/sampling.hip::sample_block). Semantically identical to
+    sample_neighbors + to_block."""
+    ext = backend.ext_for(seeds)
+    padded, counts, srcdata, counter = ext.sample_block(
+        indptr, indices, workspace.table, seeds, fanout, replace, seed
+    )
+    blk_indptr = torch.zeros(
+        counts.numel() + 1, dtype=torch.int64, device=seeds.device
+    )
+    torch.cumsum(counts, 0, out=blk_indptr[1:])
+    # one host sync fetches (n_new, total_edges) together
+    tot = torch.stack([counter[0], blk_indptr[-1]]).cpu()
+    n_new, E = int(tot[0]), int(tot[1])
+    packed = ext.pack_padded(padded, counts, blk_indptr[:-1].contiguous(), E)
+    n_seed = seeds.numel()
+    return Block(
+        blk_indptr,
+        packed,
+        num_src=n_seed + n_new,
+        num_dst=n_seed,
+        srcdata_nids=srcdata[: n_seed + n_new],
+    )
+
+
 def to_block(
     seeds: torch.Tensor,
     neighbors: torch.Tensor,
@@ -135,15 +169,21 @@ class NeighborSampler:
         cur = seeds
         self._step += 1
         for layer, fanout in enumerate(reversed(self.fanouts)):
-            nbrs, counts = sample_neighbors(
-                self.indptr,
-                self.indices,
-                cur,
-                fanout,
-                self.replace,
-                seed=(self._step * 1000003 + layer),
-            )
-            blk = to_block(cur, nbrs, counts, self.workspace)
+            if cur.is_cuda:
+                blk = sample_block_fused(
+                    self.indptr, self.indices, self.workspace, cur, fanout,
+                    self.replace, seed=(self._step * 1000003 + layer),
+                )
+            else:
+                nbrs, counts = sample_neighbors(
+                    self.indptr,
+                    self.indices,
+                    cur,
+                    fanout,
+                    self.replace,
+                    seed=(self._step * 1000003 + layer),
+                )
+                blk = to_block(cur, nbrs, counts, self.workspace)
             blocks.insert(0, blk)
             cur = blk.srcdata_nids
         return cur, seeds, blocks  # (input_nodes, output_nodes, blocks)

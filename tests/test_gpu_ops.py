@@ -255,3 +255,40 @@ def test_kge_fused_neg_matches_cpu(dev, name, neg_head):
     assert torch.allclose(h_gpu.grad.cpu(), h_cpu.grad, atol=1e-3, rtol=1e-3)
     assert torch.allclose(r_gpu.grad.cpu(), r_cpu.grad, atol=1e-3, rtol=1e-3)
     assert torch.allclose(n_gpu.grad.cpu(), n_cpu.grad, atol=1e-3, rtol=1e-3)
+
+
+def test_sample_block_fused_matches_unfused(dev, big_graph):
+    """Fused single-sync sample_block == sample_neighbors + to_block."""
+    from dgl_operator_amd.ops.sampling import (
+        CompactionWorkspace,
+        sample_block_fused,
+        sample_neighbors,
+        to_block,
+    )
+
+    g = big_graph
+    indptr, indices, _ = g.csc()
+    dip, dix = indptr.to(dev), indices.to(dev)
+    seeds = torch.randperm(g.num_nodes, device=dev)[:1500]
+    ws1 = CompactionWorkspace(g.num_nodes, dev)
+    ws2 = CompactionWorkspace(g.num_nodes, dev)
+
+    nbrs, counts = sample_neighbors(dip, dix, seeds, fanout=10, seed=42)
+    blk_a = to_block(seeds, nbrs, counts, ws1)
+    blk_b = sample_block_fused(dip, dix, ws2, seeds, fanout=10, seed=42)
+
+    assert blk_a.num_dst_nodes == blk_b.num_dst_nodes
+    assert blk_a.num_src_nodes == blk_b.num_src_nodes
+    assert torch.equal(blk_a.csc_indptr, blk_b.csc_indptr)
+    # same parent node per edge position (local ids may differ in order)
+    assert torch.equal(
+        blk_a.srcdata_nids[blk_a.csc_indices],
+        blk_b.srcdata_nids[blk_b.csc_indices],
+    )
+    # same set of source nodes; seeds first
+    assert torch.equal(blk_b.srcdata_nids[:1500], seeds)
+    assert torch.equal(
+        blk_a.srcdata_nids.sort().values, blk_b.srcdata_nids.sort().values
+    )
+    # workspace reset
+    assert int((ws2.table != -1).sum()) == 0
