@@ -23,6 +23,14 @@ namespace doa {
 
 constexpr int kWave = 64;
 
+// accumulate 2-byte float types in fp32
+template <typename T>
+struct AccT { using type = T; };
+template <>
+struct AccT<c10::Half> { using type = float; };
+template <>
+struct AccT<c10::BFloat16> { using type = float; };
+
 inline int64_t ceil_div(int64_t a, int64_t b) { return (a + b - 1) / b; }
 
 // Grid sizing: enough blocks to fill the chip several times over, but capped

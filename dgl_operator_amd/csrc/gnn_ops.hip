@@ -41,34 +41,41 @@ __global__ void spmm_kernel(
     const int f0 = c * VEC;
     const int h = (WM == W_HEAD) ? (f0 / D) : 0;
     const int64_t p0 = indptr[row], p1 = indptr[row + 1];
-    scalar_t acc[VEC];
+    using acc_t = typename AccT<scalar_t>::type;
+    acc_t acc[VEC];
 #pragma unroll
-    for (int i = 0; i < VEC; ++i) acc[i] = scalar_t(0);
+    for (int i = 0; i < VEC; ++i) acc[i] = acc_t(0);
     for (int64_t p = p0; p < p1; ++p) {
       const int64_t u = indices[p];
-      scalar_t w = scalar_t(1);
-      if (WM == W_SCALAR) w = ew[p];
-      if (WM == W_HEAD) w = ew[p * (F / D) + h];
+      acc_t w = acc_t(1);
+      if (WM == W_SCALAR) w = (acc_t)ew[p];
+      if (WM == W_HEAD) w = (acc_t)ew[p * (F / D) + h];
       const scalar_t* src = feat + u * F + f0;
       if (VEC == 4 && sizeof(scalar_t) == 4) {
         const float4 v = *reinterpret_cast<const float4*>(src);
-        acc[0] += w * ((const scalar_t*)&v)[0];
-        acc[1] += w * ((const scalar_t*)&v)[1];
-        acc[2] += w * ((const scalar_t*)&v)[2];
-        acc[3] += w * ((const scalar_t*)&v)[3];
+        acc[0] += w * (acc_t)((const scalar_t*)&v)[0];
+        acc[1] += w * (acc_t)((const scalar_t*)&v)[1];
+        acc[2] += w * (acc_t)((const scalar_t*)&v)[2];
+        acc[3] += w * (acc_t)((const scalar_t*)&v)[3];
+      } else if (VEC == 8 && sizeof(scalar_t) == 2) {
+        // 8 x bf16/half = 16 B per lane
+        const uint4 v = *reinterpret_cast<const uint4*>(src);
+        const scalar_t* e = (const scalar_t*)&v;
+#pragma unroll
+        for (int i = 0; i < 8; ++i) acc[i] += w * (acc_t)e[i];
       } else {
 #pragma unroll
-        for (int i = 0; i < VEC; ++i) acc[i] += w * src[i];
+        for (int i = 0; i < VEC; ++i) acc[i] += w * (acc_t)src[i];
       }
     }
     if (mean && p1 > p0) {
-      const scalar_t inv = scalar_t(1) / scalar_t(p1 - p0);
+      const acc_t inv = acc_t(1) / acc_t(p1 - p0);
 #pragma unroll
       for (int i = 0; i < VEC; ++i) acc[i] *= inv;
     }
     scalar_t* dst = out + row * F + f0;
 #pragma unroll
-    for (int i = 0; i < VEC; ++i) dst[i] = acc[i];
+    for (int i = 0; i < VEC; ++i) dst[i] = (scalar_t)acc[i];
   }
 }
 
@@ -86,7 +93,9 @@ static void spmm_launch(const at::Tensor& indptr, const at::Tensor& indices,
   const int block = 256;
   const bool vec4 = (F % 4 == 0) && (sizeof(scalar_t) == 4) &&
                     (wm != W_HEAD || (D % 4 == 0));
-  const int chunks = vec4 ? F / 4 : F;
+  const bool vec8 = (F % 8 == 0) && (sizeof(scalar_t) == 2) &&
+                    (wm != W_HEAD || (D % 8 == 0));
+  const int chunks = vec4 ? F / 4 : (vec8 ? F / 8 : F);
   const int grid = grid_for(num_rows * chunks, block);
   auto stream = cur_stream();
 #define DOA_SPMM(V, W)                                                        \
@@ -98,6 +107,10 @@ static void spmm_launch(const at::Tensor& indptr, const at::Tensor& indices,
     if (wm == W_NONE) DOA_SPMM(4, W_NONE);
     else if (wm == W_SCALAR) DOA_SPMM(4, W_SCALAR);
     else DOA_SPMM(4, W_HEAD);
+  } else if (vec8) {
+    if (wm == W_NONE) DOA_SPMM(8, W_NONE);
+    else if (wm == W_SCALAR) DOA_SPMM(8, W_SCALAR);
+    else DOA_SPMM(8, W_HEAD);
   } else {
     if (wm == W_NONE) DOA_SPMM(1, W_NONE);
     else if (wm == W_SCALAR) DOA_SPMM(1, W_SCALAR);
@@ -126,7 +139,7 @@ at::Tensor spmm(at::Tensor indptr, at::Tensor indices, at::Tensor feat,
   osz.push_back(num_rows);
   for (int i = 1; i < featc.dim(); ++i) osz.push_back(featc.size(i));
   auto out = at::empty(osz, featc.options());
-  AT_DISPATCH_FLOATING_TYPES(featc.scalar_type(), "spmm", [&] {
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half, featc.scalar_type(), "spmm", [&] {
     c10::optional<at::Tensor> ewc;
     if (eweight.has_value()) ewc = eweight->contiguous();
     spmm_launch<scalar_t>(indptr, indices, featc, ewc, out, F, D, mean);
@@ -149,7 +162,8 @@ __global__ void sddmm_dot_kernel(
     const int h = (int)(tid % H);
     const scalar_t* a = fu + (src[e] * H + h) * D;
     const scalar_t* b = fv + (dst[e] * H + h) * D;
-    scalar_t acc = scalar_t(0);
+    using acc_t = typename AccT<scalar_t>::type;
+    acc_t acc = acc_t(0);
     int d = 0;
     if (VEC == 4 && sizeof(scalar_t) == 4) {
       for (; d + 4 <= D; d += 4) {
@@ -158,8 +172,8 @@ __global__ void sddmm_dot_kernel(
         acc += va.x * vb.x + va.y * vb.y + va.z * vb.z + va.w * vb.w;
       }
     }
-    for (; d < D; ++d) acc += a[d] * b[d];
-    out[tid] = acc;
+    for (; d < D; ++d) acc += (acc_t)a[d] * (acc_t)b[d];
+    out[tid] = (scalar_t)acc;
   }
 }
 
@@ -176,7 +190,7 @@ at::Tensor sddmm_dot(at::Tensor src, at::Tensor dst, at::Tensor feat_u,
   const int block = 256;
   const int grid = grid_for(E * H, block);
   auto stream = cur_stream();
-  AT_DISPATCH_FLOATING_TYPES(fu.scalar_type(), "sddmm_dot", [&] {
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half, fu.scalar_type(), "sddmm_dot", [&] {
     if (D % 4 == 0 && sizeof(scalar_t) == 4) {
       hipLaunchKernelGGL((sddmm_dot_kernel<scalar_t, 4>), dim3(grid),
                          dim3(block), 0, stream, src.data_ptr<int64_t>(),
@@ -203,13 +217,15 @@ template <typename scalar_t>
 __global__ void edge_softmax_fwd_kernel(const int64_t* __restrict__ indptr,
                                         const scalar_t* __restrict__ s,
                                         scalar_t* __restrict__ out,
-                                        int64_t num_rows, int H) {
+                                        int64_t num_rows, int H,
+                                        int64_t thresh) {
   const int64_t total = num_rows * H;
   for (int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; tid < total;
        tid += (int64_t)gridDim.x * blockDim.x) {
     const int64_t row = tid / H;
     const int h = (int)(tid % H);
     const int64_t p0 = indptr[row], p1 = indptr[row + 1];
+    if (p1 - p0 > thresh) continue;  // long rows: wave kernel
     float m = -INFINITY, sum = 0.f;
     for (int64_t p = p0; p < p1; ++p) {
       const float v = (float)s[p * H + h];
@@ -233,13 +249,15 @@ __global__ void edge_softmax_bwd_kernel(const int64_t* __restrict__ indptr,
                                         const scalar_t* __restrict__ a,
                                         const scalar_t* __restrict__ g,
                                         scalar_t* __restrict__ out,
-                                        int64_t num_rows, int H) {
+                                        int64_t num_rows, int H,
+                                        int64_t thresh) {
   const int64_t total = num_rows * H;
   for (int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; tid < total;
        tid += (int64_t)gridDim.x * blockDim.x) {
     const int64_t row = tid / H;
     const int h = (int)(tid % H);
     const int64_t p0 = indptr[row], p1 = indptr[row + 1];
+    if (p1 - p0 > thresh) continue;  // long rows: wave kernel
     float acc = 0.f;
     for (int64_t p = p0; p < p1; ++p)
       acc += (float)a[p * H + h] * (float)g[p * H + h];
@@ -248,6 +266,75 @@ __global__ void edge_softmax_bwd_kernel(const int64_t* __restrict__ indptr,
           (scalar_t)((float)a[p * H + h] * ((float)g[p * H + h] - acc));
   }
 }
+
+// Wave-per-(row,head) variants for LONG segments (power-law hubs in
+// full-graph attention): lanes stride the segment, online (max, sum) pairs
+// merge across lanes with shfl_xor. The thread-per-row kernels above skip
+// rows longer than the threshold; these skip the short ones.
+template <typename scalar_t>
+__global__ void edge_softmax_fwd_long_kernel(
+    const int64_t* __restrict__ indptr, const scalar_t* __restrict__ s,
+    scalar_t* __restrict__ out, int64_t num_rows, int H, int64_t thresh) {
+  const int64_t wid = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / kWave;
+  const int lane = threadIdx.x % kWave;
+  const int64_t nwaves = (int64_t)gridDim.x * blockDim.x / kWave;
+  const int64_t total = num_rows * H;
+  for (int64_t t = wid; t < total; t += nwaves) {
+    const int64_t row = t / H;
+    const int h = (int)(t % H);
+    const int64_t p0 = indptr[row], p1 = indptr[row + 1];
+    if (p1 - p0 <= thresh) continue;
+    float m = -INFINITY, sum = 0.f;
+    for (int64_t p = p0 + lane; p < p1; p += kWave) {
+      const float v = (float)s[p * H + h];
+      if (v > m) {
+        sum = sum * __expf(m - v) + 1.f;
+        m = v;
+      } else {
+        sum += __expf(v - m);
+      }
+    }
+#pragma unroll
+    for (int off = kWave / 2; off > 0; off >>= 1) {
+      const float om = __shfl_xor(m, off, kWave);
+      const float os = __shfl_xor(sum, off, kWave);
+      const float mn = fmaxf(m, om);
+      sum = sum * __expf(m - mn) + os * __expf(om - mn);
+      m = mn;
+    }
+    const float inv = (sum > 0.f) ? 1.f / sum : 0.f;
+    for (int64_t p = p0 + lane; p < p1; p += kWave)
+      out[p * H + h] = (scalar_t)(__expf((float)s[p * H + h] - m) * inv);
+  }
+}
+
+template <typename scalar_t>
+__global__ void edge_softmax_bwd_long_kernel(
+    const int64_t* __restrict__ indptr, const scalar_t* __restrict__ a,
+    const scalar_t* __restrict__ g, scalar_t* __restrict__ out,
+    int64_t num_rows, int H, int64_t thresh) {
+  const int64_t wid = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / kWave;
+  const int lane = threadIdx.x % kWave;
+  const int64_t nwaves = (int64_t)gridDim.x * blockDim.x / kWave;
+  const int64_t total = num_rows * H;
+  for (int64_t t = wid; t < total; t += nwaves) {
+    const int64_t row = t / H;
+    const int h = (int)(t % H);
+    const int64_t p0 = indptr[row], p1 = indptr[row + 1];
+    if (p1 - p0 <= thresh) continue;
+    float acc = 0.f;
+    for (int64_t p = p0 + lane; p < p1; p += kWave)
+      acc += (float)a[p * H + h] * (float)g[p * H + h];
+#pragma unroll
+    for (int off = kWave / 2; off > 0; off >>= 1)
+      acc += __shfl_xor(acc, off, kWave);
+    for (int64_t p = p0 + lane; p < p1; p += kWave)
+      out[p * H + h] =
+          (scalar_t)((float)a[p * H + h] * ((float)g[p * H + h] - acc));
+  }
+}
+
+constexpr int64_t kSoftmaxLongRow = 64;
 
 at::Tensor edge_softmax_fwd(at::Tensor indptr, at::Tensor scores) {
   TORCH_CHECK(scores.is_cuda(), "edge_softmax: scores must be on GPU");
@@ -259,11 +346,16 @@ at::Tensor edge_softmax_fwd(at::Tensor indptr, at::Tensor scores) {
   const int block = 256;
   const int grid = grid_for(num_rows * H, block);
   auto stream = cur_stream();
-  AT_DISPATCH_FLOATING_TYPES(s.scalar_type(), "edge_softmax_fwd", [&] {
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half, s.scalar_type(), "edge_softmax_fwd", [&] {
     hipLaunchKernelGGL((edge_softmax_fwd_kernel<scalar_t>), dim3(grid),
                        dim3(block), 0, stream, indptr.data_ptr<int64_t>(),
                        s.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),
-                       num_rows, H);
+                       num_rows, H, kSoftmaxLongRow);
+    hipLaunchKernelGGL((edge_softmax_fwd_long_kernel<scalar_t>),
+                       dim3(grid_for(num_rows * H * kWave, block)),
+                       dim3(block), 0, stream, indptr.data_ptr<int64_t>(),
+                       s.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),
+                       num_rows, H, kSoftmaxLongRow);
   });
   DOA_CHECK_HIP(hipGetLastError());
   return out;
@@ -280,11 +372,18 @@ at::Tensor edge_softmax_bwd(at::Tensor indptr, at::Tensor out,
   const int block = 256;
   const int grid = grid_for(num_rows * H, block);
   auto stream = cur_stream();
-  AT_DISPATCH_FLOATING_TYPES(a.scalar_type(), "edge_softmax_bwd", [&] {
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half, a.scalar_type(), "edge_softmax_bwd", [&] {
     hipLaunchKernelGGL((edge_softmax_bwd_kernel<scalar_t>), dim3(grid),
                        dim3(block), 0, stream, indptr.data_ptr<int64_t>(),
                        a.data_ptr<scalar_t>(), g.data_ptr<scalar_t>(),
-                       gin.data_ptr<scalar_t>(), num_rows, H);
+                       gin.data_ptr<scalar_t>(), num_rows, H,
+                       kSoftmaxLongRow);
+    hipLaunchKernelGGL((edge_softmax_bwd_long_kernel<scalar_t>),
+                       dim3(grid_for(num_rows * H * kWave, block)),
+                       dim3(block), 0, stream, indptr.data_ptr<int64_t>(),
+                       a.data_ptr<scalar_t>(), g.data_ptr<scalar_t>(),
+                       gin.data_ptr<scalar_t>(), num_rows, H,
+                       kSoftmaxLongRow);
   });
   DOA_CHECK_HIP(hipGetLastError());
   return gin;
@@ -305,22 +404,23 @@ __global__ void segment_reduce_kernel(const int64_t* __restrict__ offsets,
     const int64_t seg = tid / chunks;
     const int f0 = (int)(tid % chunks) * VEC;
     const int64_t r0 = offsets[seg], r1 = offsets[seg + 1];
-    scalar_t acc[VEC];
+    using acc_t = typename AccT<scalar_t>::type;
+    acc_t acc[VEC];
 #pragma unroll
-    for (int i = 0; i < VEC; ++i) acc[i] = scalar_t(0);
+    for (int i = 0; i < VEC; ++i) acc[i] = acc_t(0);
     for (int64_t r = r0; r < r1; ++r) {
       const scalar_t* srcp = feat + r * F + f0;
 #pragma unroll
-      for (int i = 0; i < VEC; ++i) acc[i] += srcp[i];
+      for (int i = 0; i < VEC; ++i) acc[i] += (acc_t)srcp[i];
     }
     if (mean && r1 > r0) {
-      const scalar_t inv = scalar_t(1) / scalar_t(r1 - r0);
+      const acc_t inv = acc_t(1) / acc_t(r1 - r0);
 #pragma unroll
       for (int i = 0; i < VEC; ++i) acc[i] *= inv;
     }
     scalar_t* dst = out + seg * F + f0;
 #pragma unroll
-    for (int i = 0; i < VEC; ++i) dst[i] = acc[i];
+    for (int i = 0; i < VEC; ++i) dst[i] = (scalar_t)acc[i];
   }
 }
 
@@ -336,7 +436,7 @@ at::Tensor segment_reduce(at::Tensor offsets, at::Tensor feat, bool mean) {
   auto out = at::empty(osz, f.options());
   const int block = 256;
   auto stream = cur_stream();
-  AT_DISPATCH_FLOATING_TYPES(f.scalar_type(), "segment_reduce", [&] {
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half, f.scalar_type(), "segment_reduce", [&] {
     if (F % 4 == 0 && sizeof(scalar_t) == 4) {
       const int grid = grid_for(num_segs * (F / 4), block);
       hipLaunchKernelGGL((segment_reduce_kernel<scalar_t, 4>), dim3(grid),

@@ -191,3 +191,40 @@ def evaluate_kge(
         "MRR": rr_sum / n_total,
         **{f"Hits@{k}": hit_counts[k] / n_total for k in hits},
     }
+
+
+def relation_partition_order(
+    rels_of_triples: torch.Tensor, num_relations: int, num_parts: int,
+    mode: str = "soft",
+):
+    """Relation->part assignment balancing per-part TRIPLE counts — parity
+    with DGL-KE's relation partitioners (reference hotfix/sampler.py:
+    SoftRelationPartition :32-148, BalancedRelationPartition :150,
+    RandomPartition :256-290). Returns (new_of_old, boundaries): relations are
+    RELABELED so each part owns one contiguous id range (what
+    ShardedEmbedding shards by); heavy relations go first so the greedy
+    least-loaded packing balances like the reference's.
+    """
+    counts = torch.bincount(rels_of_triples, minlength=num_relations)
+    if mode == "random":
+        order = torch.randperm(
+            num_relations, generator=torch.Generator().manual_seed(0)
+        )
+        assign = torch.arange(num_relations)[order] % num_parts
+        assign = assign[torch.argsort(order)]  # assignment per original id
+    else:  # soft / balanced: greedy least-loaded over descending frequency
+        order = torch.argsort(counts, descending=True)
+        loads = [0] * num_parts
+        assign = torch.empty(num_relations, dtype=torch.int64)
+        for rid in order.tolist():
+            p = min(range(num_parts), key=lambda q: loads[q])
+            assign[rid] = p
+            loads[p] += int(counts[rid])
+    perm = torch.argsort(assign, stable=True)  # new id -> old id
+    new_of_old = torch.empty_like(perm)
+    new_of_old[perm] = torch.arange(num_relations)
+    sizes = torch.bincount(assign, minlength=num_parts)
+    boundaries = [0]
+    for s in sizes.tolist():
+        boundaries.append(boundaries[-1] + s)
+    return new_of_old, boundaries

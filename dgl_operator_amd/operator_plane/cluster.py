@@ -184,3 +184,58 @@ class FakeCluster(Cluster):
     def create_rbac(self, obj):
         self.rbac[f"{obj.namespace}/{obj.kind}/{obj.name}"] = obj
         return obj
+
+
+class KubectlCluster(Cluster):
+    """Thin kubectl-backed Cluster for running the watcher/manager against a
+    real Kubernetes API (read paths used by watcher-loop; create/update paths
+    shell out to `kubectl apply/create`). Requires a kubeconfig."""
+
+    def __init__(self, kubectl: str = "kubectl"):
+        self.kubectl = kubectl
+
+    def _get_json(self, args):
+        r = subprocess.run(
+            [self.kubectl] + args + ["-o", "json"],
+            capture_output=True, text=True,
+        )
+        if r.returncode != 0:
+            return None
+        return json.loads(r.stdout)
+
+    def get_pod(self, namespace, name):
+        d = self._get_json(["get", "pod", name, "-n", namespace])
+        if d is None:
+            return None
+        status = d.get("status", {})
+        ready = all(
+            cs.get("ready", False)
+            for cs in status.get("containerStatuses", [{"ready": False}])
+        )
+        return Pod(
+            name=name,
+            namespace=namespace,
+            labels=d.get("metadata", {}).get("labels", {}),
+            phase=PodPhase(status.get("phase", "Unknown")),
+            reason=status.get("reason"),
+            ip=status.get("podIP"),
+            containers_ready=ready,
+        )
+
+    def list_pods(self, namespace, owner):
+        d = self._get_json(
+            ["get", "pods", "-n", namespace, "-l", f"dgl-job-name={owner}"]
+        )
+        if d is None:
+            return []
+        return [
+            self.get_pod(namespace, item["metadata"]["name"])
+            for item in d.get("items", [])
+        ]
+
+    def delete_pod(self, namespace, name):
+        subprocess.run(
+            [self.kubectl, "delete", "pod", name, "-n", namespace,
+             "--ignore-not-found"],
+            capture_output=True,
+        )

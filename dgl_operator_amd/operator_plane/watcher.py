@@ -65,3 +65,36 @@ def watch(
 def watch_file(cluster: Cluster, namespace: str, watchfile_text: str,
                mode: str, **kw) -> bool:
     return watch(cluster, namespace, parse_watchfile(watchfile_text), mode, **kw)
+
+
+def main(argv=None):
+    """CLI parity with the reference watcher-loop binary
+    (/root/reference/watcher-loop/app/server.go:38-64): env NAMESPACE,
+    WATCHERFILE (a path under /etc/dgl), WATCHERMODE=ready|finished."""
+    import argparse
+    import os
+    import sys
+
+    from .cluster import KubectlCluster
+
+    p = argparse.ArgumentParser(prog="watcher-loop")
+    p.add_argument("--namespace", default=os.environ.get("NAMESPACE", "default"))
+    p.add_argument("--watcherfile",
+                   default=os.environ.get("WATCHERFILE", "/etc/dgl/hostfile"))
+    p.add_argument("--mode", default=os.environ.get("WATCHERMODE", "ready"))
+    p.add_argument("--timeout", type=float, default=None)
+    args = p.parse_args(argv)
+    path = args.watcherfile
+    if not os.path.isabs(path):
+        path = os.path.join("/etc/dgl", path)
+    with open(path) as f:
+        names = parse_watchfile(f.read())
+    print(f"[watcher-loop] waiting for {len(names)} pods to be {args.mode}",
+          flush=True)
+    ok = watch(KubectlCluster(), args.namespace, names, args.mode,
+               timeout=args.timeout)
+    sys.exit(0 if ok else 1)
+
+
+if __name__ == "__main__":
+    main()

@@ -292,3 +292,17 @@ def test_sample_block_fused_matches_unfused(dev, big_graph):
     )
     # workspace reset
     assert int((ws2.table != -1).sum()) == 0
+
+
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float16])
+def test_spmm_low_precision(dev, big_graph, dtype):
+    """bf16/fp16 SpMM (fp32 accumulation) tracks the fp32 reference."""
+    g = big_graph
+    indptr, indices, _ = g.csc()
+    x = torch.randn(g.num_nodes, 64)
+    ref = _spmm_ref(indptr, indices, x, None, True)
+    out = spmm_raw(
+        indptr.to(dev), indices.to(dev), x.to(dev).to(dtype), None, True
+    )
+    assert out.dtype == dtype
+    assert torch.allclose(out.float().cpu(), ref, atol=0.15, rtol=0.05)

@@ -136,3 +136,24 @@ def _eval_worker(rank, world):
 
 def test_kge_eval_metrics():
     _run_workers(_eval_worker)
+
+
+def test_relation_partition_balances_triples():
+    from dgl_operator_amd.distributed.kge import relation_partition_order
+
+    torch.manual_seed(3)
+    R, P = 20, 4
+    # skewed relation frequencies (one dominant)
+    rels = torch.cat([
+        torch.zeros(500, dtype=torch.int64),
+        torch.randint(1, R, (500,)),
+    ])
+    new_of_old, bounds = relation_partition_order(rels, R, P, mode="soft")
+    assert len(bounds) == P + 1 and bounds[-1] == R
+    # triple load per part under the new labeling
+    new_rels = new_of_old[rels]
+    part = torch.bucketize(new_rels, torch.tensor(bounds[1:-1]), right=True)
+    loads = torch.bincount(part, minlength=P).float()
+    # greedy packing: no part holds more than the heavy relation + slack
+    assert loads.max() <= 500 + 200
+    assert loads.min() > 0
