@@ -50,6 +50,8 @@ def parse_args():
     p.add_argument("--partition", type=str, default="range",
                    help="range|ldg partition strategy for multi-rank runs")
     p.add_argument("--device", type=str, default=None)
+    p.add_argument("--profile", type=str, default="",
+                   help="write a torch.profiler chrome trace of 3 steps here")
     return p.parse_args()
 
 
@@ -129,6 +131,18 @@ def main():
     # warmup
     for s in range(args.warmup):
         one_step(s)
+
+    if args.profile and rank == 0:
+        from torch.profiler import ProfilerActivity, profile
+
+        acts = [ProfilerActivity.CPU]
+        if device.type == "cuda":
+            acts.append(ProfilerActivity.CUDA)
+        with profile(activities=acts) as prof:
+            for s in range(3):
+                one_step(10_000 + s)
+        prof.export_chrome_trace(args.profile)
+        print(f"# wrote profiler trace to {args.profile}")
 
     if device.type == "cuda":
         torch.cuda.synchronize()

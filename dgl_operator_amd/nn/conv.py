@@ -50,8 +50,15 @@ class SAGEConv(nn.Module):
         if self.fc_neigh.bias is not None:
             nn.init.zeros_(self.fc_neigh.bias)
 
-    def forward(self, g: Union[Graph, Block], x: torch.Tensor) -> torch.Tensor:
-        h_neigh = gspmm(g, "copy_u", self.aggregator, x)
+    def forward(self, g: Union[Graph, Block], x: torch.Tensor,
+                edge_weight: torch.Tensor = None) -> torch.Tensor:
+        """Optional per-edge scalar weight gives the reference's
+        WeightedSAGEConv (u_mul_e aggregate,
+        /root/reference/examples/GraphSAGE/code/3_message_passing.py:263)."""
+        if edge_weight is None:
+            h_neigh = gspmm(g, "copy_u", self.aggregator, x)
+        else:
+            h_neigh = gspmm(g, "u_mul_e", self.aggregator, x, edge_weight)
         return self.fc_self(_dst_feat(g, x)) + self.fc_neigh(h_neigh)
 
 

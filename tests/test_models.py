@@ -106,3 +106,15 @@ def test_kge_all_score_funcs_train_step():
         negs = torch.randint(0, 30, (2, 4))
         loss = m.train_step(heads, rels, tails, negs, chunk_size=4, lr=0.05)
         assert loss == loss, name  # not NaN
+
+
+def test_weighted_sageconv():
+    from dgl_operator_amd.nn import SAGEConv
+
+    g = rmat_graph(50, 400, num_feats=6, seed=8)
+    w = torch.rand(g.num_edges, requires_grad=True)
+    layer = SAGEConv(6, 4)
+    out = layer(g, g.ndata["feat"], edge_weight=w)
+    assert out.shape == (50, 4)
+    out.sum().backward()
+    assert w.grad is not None
