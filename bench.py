@@ -110,11 +110,24 @@ def main():
     seed_gen = torch.Generator(device=device)
     seed_gen.manual_seed(12345 + rank)
     lo, hi = dg.lo, dg.hi
+    n_owned = hi - lo
+    # DataLoader semantics (and no per-step unique/sync): shuffle the owned
+    # nodes once per epoch, take contiguous distinct-seed slices
+    epoch_perm = torch.randperm(n_owned, generator=seed_gen, device=device) + lo
+    cursor = [0]
+
+    def next_seeds():
+        if cursor[0] + args.batch > n_owned:
+            epoch_perm.copy_(
+                torch.randperm(n_owned, generator=seed_gen, device=device) + lo
+            )
+            cursor[0] = 0
+        s = epoch_perm[cursor[0] : cursor[0] + args.batch]
+        cursor[0] += args.batch
+        return s
 
     def one_step(step: int) -> int:
-        seeds = torch.randint(lo, hi, (args.batch,), generator=seed_gen,
-                              device=device)
-        seeds = torch.unique(seeds)
+        seeds = next_seeds()
         input_nodes, output_nodes, blocks = dg.sample_blocks(
             seeds, fanouts, seed=step + 1
         )
@@ -122,7 +135,7 @@ def main():
         y = dg.pull("label", output_nodes)
         logits = model(blocks, x)
         loss = F.cross_entropy(logits, y)
-        opt.zero_grad(set_to_none=False)
+        opt.zero_grad(set_to_none=True)
         loss.backward()
         flat_allreduce_grads(model)
         opt.step()

@@ -6,6 +6,8 @@ namespace doa {
 
 at::Tensor spmm(at::Tensor indptr, at::Tensor indices, at::Tensor feat,
                 c10::optional<at::Tensor> eweight, bool mean);
+at::Tensor spmm_scatter(at::Tensor indptr, at::Tensor indices, at::Tensor grad,
+                        c10::optional<at::Tensor> eweight, int64_t num_src);
 at::Tensor sddmm_dot(at::Tensor src, at::Tensor dst, at::Tensor feat_u,
                      at::Tensor feat_v);
 at::Tensor edge_softmax_fwd(at::Tensor indptr, at::Tensor scores);
@@ -49,6 +51,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("spmm", &doa::spmm, "generalized SpMM (copy_u/u_mul_e x sum/mean)",
         py::arg("indptr"), py::arg("indices"), py::arg("feat"),
         py::arg("eweight") = py::none(), py::arg("mean") = false);
+  m.def("spmm_scatter", &doa::spmm_scatter,
+        "transposed SpMM via atomic scatter (block backward)",
+        py::arg("indptr"), py::arg("indices"), py::arg("grad"),
+        py::arg("eweight") = py::none(), py::arg("num_src") = 0);
   m.def("sddmm_dot", &doa::sddmm_dot, "per-edge u dot v");
   m.def("edge_softmax_fwd", &doa::edge_softmax_fwd);
   m.def("edge_softmax_bwd", &doa::edge_softmax_bwd);

@@ -148,6 +148,98 @@ at::Tensor spmm(at::Tensor indptr, at::Tensor indices, at::Tensor feat,
 }
 
 // ---------------------------------------------------------------------------
+// Transposed SpMM as scatter (backward over a Block): walks the SAME CSC
+// structure as forward and atomically accumulates grad into the source rows
+// — no per-step CSR transpose (the argsort it needs costs more than the
+// low-contention fp32 atomics on sampled blocks, where each source node is
+// referenced only a handful of times).
+//   out[u, :] += w[p] * grad[row, :]   for every csc position p with
+//                                      indices[p] == u
+// ---------------------------------------------------------------------------
+template <typename scalar_t, int VEC, WeightMode WM>
+__global__ void spmm_scatter_kernel(
+    const int64_t* __restrict__ indptr, const int64_t* __restrict__ indices,
+    const scalar_t* __restrict__ grad, const scalar_t* __restrict__ ew,
+    float* __restrict__ out, int64_t num_rows, int F, int D) {
+  const int chunks = F / VEC;
+  const int64_t total = num_rows * chunks;
+  for (int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; tid < total;
+       tid += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t row = tid / chunks;
+    const int c = (int)(tid % chunks);
+    const int f0 = c * VEC;
+    const int h = (WM == W_HEAD) ? (f0 / D) : 0;
+    const int64_t p0 = indptr[row], p1 = indptr[row + 1];
+    float g[VEC];
+    const scalar_t* grow = grad + row * F + f0;
+#pragma unroll
+    for (int i = 0; i < VEC; ++i) g[i] = (float)grow[i];
+    for (int64_t p = p0; p < p1; ++p) {
+      const int64_t u = indices[p];
+      float w = 1.f;
+      if (WM == W_SCALAR) w = (float)ew[p];
+      if (WM == W_HEAD) w = (float)ew[p * (F / D) + h];
+      float* dst = out + u * F + f0;
+#pragma unroll
+      for (int i = 0; i < VEC; ++i) atomicAdd(&dst[i], w * g[i]);
+    }
+  }
+}
+
+at::Tensor spmm_scatter(at::Tensor indptr, at::Tensor indices, at::Tensor grad,
+                        c10::optional<at::Tensor> eweight, int64_t num_src) {
+  TORCH_CHECK(grad.is_cuda(), "spmm_scatter: grad must be on GPU");
+  auto gradc = grad.contiguous();
+  const int64_t num_rows = indptr.numel() - 1;
+  int F = 1, D = 1;
+  for (int i = 1; i < gradc.dim(); ++i) F *= gradc.size(i);
+  D = F;
+  if (eweight.has_value() && eweight->dim() == 2) {
+    const int H = eweight->size(1);
+    D = F / H;
+  }
+  std::vector<int64_t> osz;
+  osz.push_back(num_src);
+  for (int i = 1; i < gradc.dim(); ++i) osz.push_back(gradc.size(i));
+  // fp32 accumulator (atomicAdd); cast to grad dtype at the end
+  auto out = at::zeros(osz, gradc.options().dtype(at::kFloat));
+  const int block = 256;
+  auto stream = cur_stream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half, gradc.scalar_type(), "spmm_scatter", [&] {
+    WeightMode wm = W_NONE;
+    const scalar_t* ewp = nullptr;
+    c10::optional<at::Tensor> ewc;
+    if (eweight.has_value()) {
+      ewc = eweight->contiguous();
+      ewp = ewc->data_ptr<scalar_t>();
+      wm = (ewc->dim() == 2) ? W_HEAD : W_SCALAR;
+    }
+    const bool vec4 = (F % 4 == 0) && (wm != W_HEAD || (D % 4 == 0));
+    const int chunks = vec4 ? F / 4 : F;
+    const int grid = grid_for(num_rows * chunks, block);
+#define DOA_SCAT(V, W)                                                        \
+    hipLaunchKernelGGL((spmm_scatter_kernel<scalar_t, V, W>), dim3(grid),     \
+                       dim3(block), 0, stream, indptr.data_ptr<int64_t>(),    \
+                       indices.data_ptr<int64_t>(),                           \
+                       gradc.data_ptr<scalar_t>(), ewp,                       \
+                       out.data_ptr<float>(), num_rows, F, D)
+    if (vec4) {
+      if (wm == W_NONE) DOA_SCAT(4, W_NONE);
+      else if (wm == W_SCALAR) DOA_SCAT(4, W_SCALAR);
+      else DOA_SCAT(4, W_HEAD);
+    } else {
+      if (wm == W_NONE) DOA_SCAT(1, W_NONE);
+      else if (wm == W_SCALAR) DOA_SCAT(1, W_SCALAR);
+      else DOA_SCAT(1, W_HEAD);
+    }
+#undef DOA_SCAT
+  });
+  DOA_CHECK_HIP(hipGetLastError());
+  return out.scalar_type() == gradc.scalar_type() ? out
+                                                  : out.to(gradc.scalar_type());
+}
+
+// ---------------------------------------------------------------------------
 // SDDMM u_dot_v: out[e, h] = sum_d fu[src[e], h, d] * fv[dst[e], h, d]
 // ---------------------------------------------------------------------------
 template <typename scalar_t, int VEC>

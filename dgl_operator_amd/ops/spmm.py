@@ -98,6 +98,21 @@ class _GSpMM(torch.autograd.Function):
         # transpose structure: per-src positions; csr_eids maps csr position ->
         # csc position is NOT direct: both eids map into original edge order.
         cindptr, cindices, ceids = gstruct.csc()
+        if ctx.needs_input_grad[1] and grad_out.is_cuda:
+            # GPU: scatter-atomic transposed SpMM over the SAME CSC structure
+            # (csrc spmm_scatter) — avoids building a per-block CSR (argsort)
+            # every backward; weights are already in csc order.
+            ext = backend.ext_for(grad_out)
+            grad_feat = ext.spmm_scatter(
+                cindptr, cindices, grad_out,
+                eweight if has_w else None, feat.shape[0],
+            )
+            if has_w and ctx.needs_input_grad[2]:
+                from .sddmm import sddmm_dot_raw
+
+                dst = _edge_dst(cindptr)
+                grad_w = sddmm_dot_raw(cindices, dst, feat, grad_out)
+            return None, grad_feat, grad_w, None
         rindptr, rindices, reids = gstruct.csr()
         if ctx.needs_input_grad[1]:
             w_r = None
