@@ -24,7 +24,8 @@ std::tuple<at::Tensor, at::Tensor> compact_ids(at::Tensor table,
                                                at::Tensor neighbors);
 std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> sample_block(
     at::Tensor indptr, at::Tensor indices, at::Tensor table, at::Tensor seeds,
-    int64_t fanout, bool replace, int64_t seed);
+    int64_t fanout, bool replace, int64_t seed,
+    c10::optional<at::Tensor> seed_dev);
 at::Tensor pack_padded(at::Tensor padded, at::Tensor counts,
                        at::Tensor offsets, int64_t total);
 void sparse_adagrad(at::Tensor emb, at::Tensor state, at::Tensor ids,
@@ -61,7 +62,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("segment_reduce", &doa::segment_reduce);
   m.def("sample_neighbors", &doa::sample_neighbors);
   m.def("compact_ids", &doa::compact_ids);
-  m.def("sample_block", &doa::sample_block);
+  m.def("sample_block", &doa::sample_block, py::arg("indptr"),
+        py::arg("indices"), py::arg("table"), py::arg("seeds"),
+        py::arg("fanout"), py::arg("replace"), py::arg("seed"),
+        py::arg("seed_dev") = py::none());
   m.def("pack_padded", &doa::pack_padded);
   m.def("sparse_adagrad", &doa::sparse_adagrad);
   m.def("ldg_partition", &doa::ldg_partition);

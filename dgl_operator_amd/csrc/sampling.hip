@@ -21,7 +21,14 @@ __global__ void sample_kernel(const int64_t* __restrict__ indptr,
                               const int64_t* __restrict__ seeds,
                               int64_t* __restrict__ out,   // [n, fanout] padded
                               int64_t* __restrict__ counts,  // [n]
-                              int64_t n, int fanout, uint64_t rngseed) {
+                              int64_t n, int fanout, uint64_t rngseed_scalar,
+                              const int64_t* __restrict__ rngseed_dev) {
+  // hipGraph-replayable RNG: when a device seed buffer is given, mix it in —
+  // the host updates the buffer between replays, so a captured graph draws
+  // fresh samples every step.
+  const uint64_t rngseed =
+      rngseed_dev ? (rngseed_scalar ^ (uint64_t)rngseed_dev[0])
+                  : rngseed_scalar;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     const int64_t v = seeds[i];
@@ -80,13 +87,13 @@ std::tuple<at::Tensor, at::Tensor> sample_neighbors(at::Tensor indptr,
                        stream, indptr.data_ptr<int64_t>(),
                        indices.data_ptr<int64_t>(), seeds.data_ptr<int64_t>(),
                        padded.data_ptr<int64_t>(), counts.data_ptr<int64_t>(),
-                       n, (int)fanout, (uint64_t)seed);
+                       n, (int)fanout, (uint64_t)seed, (const int64_t*)nullptr);
   } else {
     hipLaunchKernelGGL((sample_kernel<false>), dim3(grid), dim3(block), 0,
                        stream, indptr.data_ptr<int64_t>(),
                        indices.data_ptr<int64_t>(), seeds.data_ptr<int64_t>(),
                        padded.data_ptr<int64_t>(), counts.data_ptr<int64_t>(),
-                       n, (int)fanout, (uint64_t)seed);
+                       n, (int)fanout, (uint64_t)seed, (const int64_t*)nullptr);
   }
   DOA_CHECK_HIP(hipGetLastError());
   // pack the padded matrix into a flat neighbor list
@@ -285,13 +292,18 @@ __global__ void pack_padded_kernel(const int64_t* __restrict__ padded,
 //          totals [1] = n_new)  — all device tensors, no host sync here.
 std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> sample_block(
     at::Tensor indptr, at::Tensor indices, at::Tensor table, at::Tensor seeds,
-    int64_t fanout, bool replace, int64_t seed) {
+    int64_t fanout, bool replace, int64_t seed,
+    c10::optional<at::Tensor> seed_dev) {
   TORCH_CHECK(seeds.is_cuda(), "sample_block: GPU tensors expected");
   const int64_t n = seeds.numel();
   auto padded = at::empty({n, fanout}, seeds.options());
   auto counts = at::empty({n}, seeds.options());
-  auto srcdata = at::empty({n + n * fanout}, seeds.options());
+  // zero-filled so the unclaimed tail is a VALID node id (0) — capture mode
+  // consumes the worst-case buffer without reading the claim counter
+  auto srcdata = at::zeros({n + n * fanout}, seeds.options());
   auto counter = at::zeros({1}, seeds.options());
+  const int64_t* sdev =
+      seed_dev.has_value() ? seed_dev->data_ptr<int64_t>() : nullptr;
   const int block = 256;
   auto stream = c10::hip::getCurrentHIPStream().stream();
   if (replace) {
@@ -299,13 +311,13 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> sample_block(
                        dim3(block), 0, stream, indptr.data_ptr<int64_t>(),
                        indices.data_ptr<int64_t>(), seeds.data_ptr<int64_t>(),
                        padded.data_ptr<int64_t>(), counts.data_ptr<int64_t>(),
-                       n, (int)fanout, (uint64_t)seed);
+                       n, (int)fanout, (uint64_t)seed, sdev);
   } else {
     hipLaunchKernelGGL((sample_kernel<false>), dim3(grid_for(n, block)),
                        dim3(block), 0, stream, indptr.data_ptr<int64_t>(),
                        indices.data_ptr<int64_t>(), seeds.data_ptr<int64_t>(),
                        padded.data_ptr<int64_t>(), counts.data_ptr<int64_t>(),
-                       n, (int)fanout, (uint64_t)seed);
+                       n, (int)fanout, (uint64_t)seed, sdev);
   }
   hipLaunchKernelGGL(compact_seed_kernel, dim3(grid_for(n, block)),
                      dim3(block), 0, stream, table.data_ptr<int64_t>(),

@@ -100,13 +100,15 @@ def sample_block_fused(
     fanout: int,
     replace: bool = False,
     seed: int = 0,
+    seed_dev: "torch.Tensor | None" = None,
 ) -> Block:
     """GPU fast path: fused sample+compact with ONE host sync per hop
     (csrc/sampling.hip::sample_block). Semantically identical to
     sample_neighbors + to_block."""
     ext = backend.ext_for(seeds)
     padded, counts, srcdata, counter = ext.sample_block(
-        indptr, indices, workspace.table, seeds, fanout, replace, seed
+        indptr, indices, workspace.table, seeds, fanout, replace, seed,
+        seed_dev,
     )
     blk_indptr = torch.zeros(
         counts.numel() + 1, dtype=torch.int64, device=seeds.device
@@ -143,6 +145,43 @@ def to_block(
         num_dst=seeds.numel(),
         srcdata_nids=srcdata_nids,
     )
+
+
+def sample_block_capture(
+    indptr: torch.Tensor,
+    indices: torch.Tensor,
+    workspace: CompactionWorkspace,
+    seeds: torch.Tensor,
+    fanout: int,
+    seed: int,
+    seed_dev: torch.Tensor,
+) -> "tuple[Block, torch.Tensor]":
+    """hipGraph-capturable block build: NO host synchronization. Shapes are
+    worst-case (num_src = n + n*fanout, packed buffer n*fanout); the actual
+    sizes live on device (the claim counter / indptr[-1]) and every consumer
+    (SpMM, pull-gather) only reads the valid prefix, because the unclaimed
+    srcdata tail is node id 0 (valid) and SpMM walks indptr. RNG comes from
+    ``seed_dev`` (updated between replays) mixed with the static ``seed``.
+
+    Returns (block, counter) — counter[0] is the device-side count of newly
+    claimed source nodes (needed for the valid-edge metric)."""
+    ext = backend.ext_for(seeds)
+    padded, counts, srcdata, counter = ext.sample_block(
+        indptr, indices, workspace.table, seeds, fanout, False, seed, seed_dev
+    )
+    n = seeds.numel()
+    blk_indptr = torch.zeros(n + 1, dtype=torch.int64, device=seeds.device)
+    torch.cumsum(counts, 0, out=blk_indptr[1:])
+    packed = ext.pack_padded(padded, counts, blk_indptr[:-1].contiguous(),
+                             n * fanout)
+    blk = Block(
+        blk_indptr,
+        packed,
+        num_src=n + n * fanout,  # worst case; valid prefix tracked on device
+        num_dst=n,
+        srcdata_nids=srcdata,
+    )
+    return blk, counter
 
 
 class NeighborSampler:
