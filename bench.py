@@ -80,6 +80,8 @@ def main():
     from dgl_operator_amd.models import GraphSAGE
 
     rank, ws = comm.init_from_env()
+    if ws > 1 and args.gpus != ws and rank == 0:
+        print(f"# note: --gpus {args.gpus} overridden by WORLD_SIZE={ws}")
     if args.device:
         device = torch.device(args.device)
     elif torch.cuda.is_available():

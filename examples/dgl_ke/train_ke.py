@@ -39,6 +39,9 @@ def main():
     p.add_argument("--num-triples", type=int, default=5_000_000)
     p.add_argument("--save-path", default="")
     p.add_argument("--no-save-emb", action="store_true")
+    p.add_argument("--eval", action="store_true",
+                   help="report raw MRR/MR/Hits@K on held-out triples")
+    p.add_argument("--num-eval", type=int, default=500)
     args = p.parse_args()
 
     from dgl_operator_amd.distributed import DistKGEModel, KGEdgeSampler, comm
@@ -79,6 +82,13 @@ def main():
             tps = step * args.batch_size * ws / elapsed
             print(f"step {step} loss {loss:.4f} {tps:,.0f} triples/s",
                   flush=True)
+    if args.eval:
+        from dgl_operator_amd.distributed.kge import evaluate_kge
+
+        m = evaluate_kge(model, h[: args.num_eval], r[: args.num_eval],
+                         t[: args.num_eval])
+        if rank == 0:
+            print("eval:", {k: round(v, 4) for k, v in m.items()}, flush=True)
     if args.save_path and not args.no_save_emb:
         os.makedirs(args.save_path, exist_ok=True)
         model.entities.save_shard(

@@ -132,8 +132,16 @@ def main():
             t_bwd = time.time()
             if step % args.log_every == 0 and rank == 0:
                 n_edges = sum(b.num_edges for b in blocks)
+                speed = seeds.numel() / max(t_bwd - tic, 1e-9)
+                gpu_mb = (
+                    torch.cuda.max_memory_allocated() / 1e6
+                    if device.type == "cuda" else 0.0
+                )
+                # reference per-step format parity (train_dist.py:246-250):
+                # Speed (samples/sec) + GPU MB + phase split
                 print(
                     f"Epoch {epoch:03d} | Step {step:05d} | Loss {loss:.4f} | "
+                    f"Speed (samples/sec) {speed:.1f} | GPU {gpu_mb:.1f} MB | "
                     f"Edges {n_edges} | sample {t_sample - tic:.3f}s "
                     f"fwd {t_fwd - t_sample:.3f}s bwd+upd {t_bwd - t_fwd:.3f}s",
                     flush=True,

@@ -293,3 +293,17 @@ def test_manager_loop_and_health(tmp_path):
         assert mgr.get("default", "graphsage-dist") is None
     finally:
         mgr.stop()
+
+
+def test_shipped_manifests_parse():
+    import glob
+    import os
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    manifests = glob.glob(os.path.join(repo, "examples/v1alpha1/*.yaml"))
+    assert len(manifests) >= 3
+    for path in manifests:
+        with open(path) as f:
+            job = job_from_manifest(f.read())
+        assert job.name
+        assert job.num_workers() >= 1
