@@ -81,8 +81,24 @@ def build(force: bool = False, verbose: bool = True) -> Path:
             raise RuntimeError(f"hipcc failed on {src_name}:\n{r.stdout}\n{r.stderr}")
         return obj
 
+    failures = []
+
+    def compile_safe(name):
+        try:
+            return compile_one(name)
+        except Exception as e:  # noqa: BLE001
+            failures.append(str(e))
+            return None
+
     with ThreadPoolExecutor(max_workers=8) as pool:
-        objs = [o for o in pool.map(compile_one, SOURCES) if o is not None]
+        objs = [o for o in pool.map(compile_safe, SOURCES) if o is not None]
+    if failures:
+        raise RuntimeError(
+            "native build FAILED:\n" + "\n\n".join(failures)
+        )
+    missing = [s for s in SOURCES if (CSRC / s).exists()] 
+    if len(objs) != len(missing):
+        raise RuntimeError("native build: object count mismatch")
 
     if force or not out_so.exists() or any(
         o.stat().st_mtime > out_so.stat().st_mtime for o in objs

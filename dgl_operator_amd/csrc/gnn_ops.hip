@@ -186,11 +186,49 @@ __global__ void spmm_scatter_kernel(
   }
 }
 
+template <typename scalar_t>
+static void spmm_scatter_launch(const at::Tensor& indptr,
+                                const at::Tensor& indices,
+                                const at::Tensor& gradc,
+                                const c10::optional<at::Tensor>& eweight,
+                                at::Tensor& out, int F, int D) {
+  const int64_t num_rows = indptr.numel() - 1;
+  WeightMode wm = W_NONE;
+  const scalar_t* ewp = nullptr;
+  c10::optional<at::Tensor> ewc;
+  if (eweight.has_value()) {
+    ewc = eweight->contiguous();
+    ewp = ewc->data_ptr<scalar_t>();
+    wm = (ewc->dim() == 2) ? W_HEAD : W_SCALAR;
+  }
+  const int block = 256;
+  const bool vec4 = (F % 4 == 0) && (wm != W_HEAD || (D % 4 == 0));
+  const int chunks = vec4 ? F / 4 : F;
+  const int grid = grid_for(num_rows * chunks, block);
+  auto stream = cur_stream();
+#define DOA_SCAT(V, W)                                                        \
+  hipLaunchKernelGGL((spmm_scatter_kernel<scalar_t, V, W>), dim3(grid),       \
+                     dim3(block), 0, stream, indptr.data_ptr<int64_t>(),      \
+                     indices.data_ptr<int64_t>(),                             \
+                     gradc.data_ptr<scalar_t>(), ewp,                         \
+                     out.data_ptr<float>(), num_rows, F, D)
+  if (vec4) {
+    if (wm == W_NONE) DOA_SCAT(4, W_NONE);
+    else if (wm == W_SCALAR) DOA_SCAT(4, W_SCALAR);
+    else DOA_SCAT(4, W_HEAD);
+  } else {
+    if (wm == W_NONE) DOA_SCAT(1, W_NONE);
+    else if (wm == W_SCALAR) DOA_SCAT(1, W_SCALAR);
+    else DOA_SCAT(1, W_HEAD);
+  }
+#undef DOA_SCAT
+  DOA_CHECK_HIP(hipGetLastError());
+}
+
 at::Tensor spmm_scatter(at::Tensor indptr, at::Tensor indices, at::Tensor grad,
                         c10::optional<at::Tensor> eweight, int64_t num_src) {
   TORCH_CHECK(grad.is_cuda(), "spmm_scatter: grad must be on GPU");
   auto gradc = grad.contiguous();
-  const int64_t num_rows = indptr.numel() - 1;
   int F = 1, D = 1;
   for (int i = 1; i < gradc.dim(); ++i) F *= gradc.size(i);
   D = F;
@@ -203,38 +241,9 @@ at::Tensor spmm_scatter(at::Tensor indptr, at::Tensor indices, at::Tensor grad,
   for (int i = 1; i < gradc.dim(); ++i) osz.push_back(gradc.size(i));
   // fp32 accumulator (atomicAdd); cast to grad dtype at the end
   auto out = at::zeros(osz, gradc.options().dtype(at::kFloat));
-  const int block = 256;
-  auto stream = cur_stream();
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half, gradc.scalar_type(), "spmm_scatter", [&] {
-    WeightMode wm = W_NONE;
-    const scalar_t* ewp = nullptr;
-    c10::optional<at::Tensor> ewc;
-    if (eweight.has_value()) {
-      ewc = eweight->contiguous();
-      ewp = ewc->data_ptr<scalar_t>();
-      wm = (ewc->dim() == 2) ? W_HEAD : W_SCALAR;
-    }
-    const bool vec4 = (F % 4 == 0) && (wm != W_HEAD || (D % 4 == 0));
-    const int chunks = vec4 ? F / 4 : F;
-    const int grid = grid_for(num_rows * chunks, block);
-#define DOA_SCAT(V, W)                                                        \
-    hipLaunchKernelGGL((spmm_scatter_kernel<scalar_t, V, W>), dim3(grid),     \
-                       dim3(block), 0, stream, indptr.data_ptr<int64_t>(),    \
-                       indices.data_ptr<int64_t>(),                           \
-                       gradc.data_ptr<scalar_t>(), ewp,                       \
-                       out.data_ptr<float>(), num_rows, F, D)
-    if (vec4) {
-      if (wm == W_NONE) DOA_SCAT(4, W_NONE);
-      else if (wm == W_SCALAR) DOA_SCAT(4, W_SCALAR);
-      else DOA_SCAT(4, W_HEAD);
-    } else {
-      if (wm == W_NONE) DOA_SCAT(1, W_NONE);
-      else if (wm == W_SCALAR) DOA_SCAT(1, W_SCALAR);
-      else DOA_SCAT(1, W_HEAD);
-    }
-#undef DOA_SCAT
+    spmm_scatter_launch<scalar_t>(indptr, indices, gradc, eweight, out, F, D);
   });
-  DOA_CHECK_HIP(hipGetLastError());
   return out.scalar_type() == gradc.scalar_type() ? out
                                                   : out.to(gradc.scalar_type());
 }
