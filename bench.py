@@ -49,6 +49,9 @@ def parse_args():
     p.add_argument("--dropout", type=float, default=0.5)
     p.add_argument("--partition", type=str, default="range",
                    help="range|ldg partition strategy for multi-rank runs")
+    p.add_argument("--no-halo", action="store_true",
+                   help="disable ghost-zone replication (fall back to "
+                        "per-step alltoallv sampling + feature pulls)")
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--profile", type=str, default="",
                    help="write a torch.profiler chrome trace of 3 steps here")
@@ -103,6 +106,10 @@ def main():
     dg = DistGraph.from_full_graph(g, book, rank)
     # free the full graph copies we no longer need (features stay sharded)
     del g
+    if ws > 1 and not args.no_halo:
+        # one-time ghost-zone replication: per-step sampling + feature pulls
+        # become communication-free (only the gradient all-reduce remains)
+        dg.build_halo_cache(args.layers, feat_keys=("feat", "label"))
 
     model = GraphSAGE(args.feat, args.hidden, args.classes,
                       n_layers=args.layers, dropout=args.dropout).to(device)
@@ -273,7 +280,11 @@ def main():
                 "model": f"GraphSAGE L{args.layers} h{args.hidden} fanout[{args.fanout}] dropout{args.dropout}",
                 "global_batch": args.batch * ws,
                 "seq_len": None,
-                "parallelism": f"dp{ws} (graph-partition data parallel, alltoallv halo pulls)",
+                "parallelism": (
+                    f"dp{ws} (graph-partition data parallel, "
+                    + ("ghost-zone halo replication)" if ws > 1 and not args.no_halo
+                       else "alltoallv halo pulls)")
+                ),
             },
         }))
 

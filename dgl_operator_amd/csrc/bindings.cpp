@@ -25,7 +25,7 @@ std::tuple<at::Tensor, at::Tensor> compact_ids(at::Tensor table,
 std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> sample_block(
     at::Tensor indptr, at::Tensor indices, at::Tensor table, at::Tensor seeds,
     int64_t fanout, bool replace, int64_t seed,
-    c10::optional<at::Tensor> seed_dev);
+    c10::optional<at::Tensor> seed_dev, c10::optional<at::Tensor> rows_opt);
 at::Tensor pack_padded(at::Tensor padded, at::Tensor counts,
                        at::Tensor offsets, int64_t total);
 void sparse_adagrad(at::Tensor emb, at::Tensor state, at::Tensor ids,
@@ -65,7 +65,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sample_block", &doa::sample_block, py::arg("indptr"),
         py::arg("indices"), py::arg("table"), py::arg("seeds"),
         py::arg("fanout"), py::arg("replace"), py::arg("seed"),
-        py::arg("seed_dev") = py::none());
+        py::arg("seed_dev") = py::none(), py::arg("rows") = py::none());
   m.def("pack_padded", &doa::pack_padded);
   m.def("sparse_adagrad", &doa::sparse_adagrad);
   m.def("ldg_partition", &doa::ldg_partition);

@@ -18,7 +18,7 @@ namespace doa {
 template <bool REPLACE>
 __global__ void sample_kernel(const int64_t* __restrict__ indptr,
                               const int64_t* __restrict__ indices,
-                              const int64_t* __restrict__ seeds,
+                              const int64_t* __restrict__ rows,  // CSC row ids
                               int64_t* __restrict__ out,   // [n, fanout] padded
                               int64_t* __restrict__ counts,  // [n]
                               int64_t n, int fanout, uint64_t rngseed_scalar,
@@ -31,7 +31,7 @@ __global__ void sample_kernel(const int64_t* __restrict__ indptr,
                   : rngseed_scalar;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
-    const int64_t v = seeds[i];
+    const int64_t v = rows[i];
     const int64_t p0 = indptr[v];
     const int64_t deg = indptr[v + 1] - p0;
     int64_t* mine = out + i * fanout;
@@ -50,8 +50,10 @@ __global__ void sample_kernel(const int64_t* __restrict__ indptr,
         for (int64_t j = 0; j < deg; ++j) mine[j] = indices[p0 + j];
         counts[i] = deg;
       } else {
-        // Floyd's sampling of `fanout` distinct positions in [0, deg)
-        int64_t sel[256];
+        // Floyd's sampling of `fanout` distinct positions in [0, deg).
+        // Selected positions live in the output row itself (global memory,
+        // L1-hot, <= fanout entries) — a runtime-indexed local array would
+        // be allocated in scratch.
         int cnt = 0;
         for (int64_t j = deg - fanout; j < deg; ++j) {
           const uint64_t t =
@@ -59,10 +61,10 @@ __global__ void sample_kernel(const int64_t* __restrict__ indptr,
                          (uint64_t)(j + 1));
           bool seen = false;
           for (int k = 0; k < cnt; ++k)
-            if (sel[k] == (int64_t)t) { seen = true; break; }
-          sel[cnt++] = seen ? j : (int64_t)t;
+            if (mine[k] == (int64_t)t) { seen = true; break; }
+          mine[cnt++] = seen ? j : (int64_t)t;
         }
-        for (int k = 0; k < cnt; ++k) mine[k] = indices[p0 + sel[k]];
+        for (int k = 0; k < cnt; ++k) mine[k] = indices[p0 + mine[k]];
         counts[i] = cnt;
       }
     }
@@ -293,8 +295,11 @@ __global__ void pack_padded_kernel(const int64_t* __restrict__ padded,
 std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> sample_block(
     at::Tensor indptr, at::Tensor indices, at::Tensor table, at::Tensor seeds,
     int64_t fanout, bool replace, int64_t seed,
-    c10::optional<at::Tensor> seed_dev) {
+    c10::optional<at::Tensor> seed_dev, c10::optional<at::Tensor> rows_opt) {
   TORCH_CHECK(seeds.is_cuda(), "sample_block: GPU tensors expected");
+  // `seeds` are GLOBAL node ids (table/srcdata keys); `rows` are their CSC
+  // row indices — identical unless sampling a halo-extended local structure.
+  at::Tensor rows = rows_opt.has_value() ? *rows_opt : seeds;
   const int64_t n = seeds.numel();
   auto padded = at::empty({n, fanout}, seeds.options());
   auto counts = at::empty({n}, seeds.options());
@@ -309,13 +314,13 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> sample_block(
   if (replace) {
     hipLaunchKernelGGL((sample_kernel<true>), dim3(grid_for(n, block)),
                        dim3(block), 0, stream, indptr.data_ptr<int64_t>(),
-                       indices.data_ptr<int64_t>(), seeds.data_ptr<int64_t>(),
+                       indices.data_ptr<int64_t>(), rows.data_ptr<int64_t>(),
                        padded.data_ptr<int64_t>(), counts.data_ptr<int64_t>(),
                        n, (int)fanout, (uint64_t)seed, sdev);
   } else {
     hipLaunchKernelGGL((sample_kernel<false>), dim3(grid_for(n, block)),
                        dim3(block), 0, stream, indptr.data_ptr<int64_t>(),
-                       indices.data_ptr<int64_t>(), seeds.data_ptr<int64_t>(),
+                       indices.data_ptr<int64_t>(), rows.data_ptr<int64_t>(),
                        padded.data_ptr<int64_t>(), counts.data_ptr<int64_t>(),
                        n, (int)fanout, (uint64_t)seed, sdev);
   }

@@ -101,6 +101,7 @@ def sample_block_fused(
     replace: bool = False,
     seed: int = 0,
     seed_dev: "torch.Tensor | None" = None,
+    rows: "torch.Tensor | None" = None,
 ) -> Block:
     """GPU fast path: fused sample+compact with ONE host sync per hop
     (csrc/sampling.hip::sample_block). Semantically identical to
@@ -108,7 +109,7 @@ def sample_block_fused(
     ext = backend.ext_for(seeds)
     padded, counts, srcdata, counter = ext.sample_block(
         indptr, indices, workspace.table, seeds, fanout, replace, seed,
-        seed_dev,
+        seed_dev, rows,
     )
     blk_indptr = torch.zeros(
         counts.numel() + 1, dtype=torch.int64, device=seeds.device
