@@ -43,6 +43,9 @@ def parse_args():
     p.add_argument("--eval-every", type=int, default=0)
     p.add_argument("--checkpoint-path", type=str, default="",
                    help="save model+optimizer per epoch; resume if present")
+    p.add_argument("--no-halo", action="store_true",
+                   help="disable ghost-zone replication (per-step alltoallv "
+                        "pulls instead)")
     return p.parse_args()
 
 
@@ -61,6 +64,8 @@ def main():
     # with slots=1 this is rank)
     part_id = int(os.environ.get("GROUP_RANK", os.environ.get("RANK", rank)))
     dg = DistGraph.from_partition(args.part_config, part_id, device=device)
+    if ws > 1 and not args.no_halo:
+        dg.build_halo_cache(args.num_layers, feat_keys=("feat", "label"))
 
     from dgl_operator_amd.models import GraphSAGE
 
