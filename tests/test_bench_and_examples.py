@@ -1,0 +1,90 @@
+"""Driver-contract guard: bench.py emits the required JSON line; example
+workloads run end-to-end with tiny configs; dglkerun drives a 2-node KE run
+over the local fabric."""
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _run(cmd, timeout=240, env=None, cwd=REPO):
+    full_env = dict(os.environ)
+    full_env.update(env or {})
+    return subprocess.run(cmd, capture_output=True, text=True, timeout=timeout,
+                          env=full_env, cwd=cwd)
+
+
+@pytest.mark.timeout(300)
+def test_bench_json_contract():
+    r = _run([sys.executable, "bench.py", "--steps", "2", "--warmup", "1",
+              "--nodes", "20000", "--edges", "100000"])
+    assert r.returncode == 0, r.stderr
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    d = json.loads(line)
+    for key in ["metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"]:
+        assert key in d, key
+    assert d["n_gpus"] == 1
+    assert d["higher_is_better"] is True
+    assert d["scaling"] == "weak"
+    assert d["dtype"] == "fp32"
+    assert d["value"] > 0
+    assert "edges/sec" in d["metric"]
+    assert "global_batch" in d["config"] and "parallelism" in d["config"]
+
+
+@pytest.mark.timeout(240)
+@pytest.mark.parametrize("script,args", [
+    ("examples/node_classification/train.py", ["--epochs", "3", "--feat", "32"]),
+    ("examples/link_predict/train.py",
+     ["--epochs", "3", "--nodes", "300", "--edges", "2000"]),
+    ("examples/graph_classification/train.py",
+     ["--epochs", "2", "--num-graphs", "20"]),
+    ("examples/dgl_ke/train_ke.py",
+     ["--max-step", "10", "--log-interval", "5", "--num-entities", "2000",
+      "--num-relations", "10", "--num-triples", "5000", "--hidden-dim", "16",
+      "--eval", "--num-eval", "50"]),
+])
+def test_example_workloads_run(script, args):
+    r = _run([sys.executable, os.path.join(REPO, script)] + args)
+    assert r.returncode == 0, f"{script}: {r.stderr[-2000:]}"
+
+
+@pytest.mark.timeout(300)
+def test_dglkerun_end_to_end(tmp_path):
+    hostfile = tmp_path / "hostfile"
+    hostfile.write_text(
+        "127.0.0.1 30050 ke-worker-0 slots=1\n"
+        "127.0.0.1 30050 ke-worker-1 slots=1\n"
+    )
+    import socket
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    r = _run(
+        [sys.executable, "-m", "dgl_operator_amd.tools.dglkerun",
+         "--hostfile", str(hostfile),
+         "--workspace", "ws",
+         "--master-port", str(port),
+         "--train-entry-point", os.path.join(REPO, "examples/dgl_ke/train_ke.py"),
+         "--model-name", "TransE_l2", "--hidden-dim", "16",
+         "--batch-size", "64", "--neg-sample-size", "8", "--max-step", "20",
+         "--save-path", "ckpts"],
+        env={"DGL_LOCAL_FABRIC_ROOT": str(tmp_path / "pods"),
+             "PYTHONPATH": REPO},
+        cwd=str(tmp_path),
+    )
+    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+    assert "Phase 5/5 dglke train" in r.stdout
+    # sharded checkpoints written per pod per rank
+    for i in range(2):
+        pd = tmp_path / "pods" / f"ke-worker-{i}"
+        found = list(pd.rglob("entity_shard*.pt"))
+        assert found, f"no entity shard under {pd}"

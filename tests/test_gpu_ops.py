@@ -337,3 +337,17 @@ def test_spmm_backward_scatter_matches_cpu(dev, big_graph, weighted):
     if weighted:
         assert torch.allclose(w_gpu.grad.cpu(), w_cpu.grad, atol=1e-3,
                               rtol=1e-3)
+
+
+def test_compact_ids_heavy_duplicates(dev):
+    """Atomic-claim stress: every neighbor is one of 3 hot ids."""
+    from dgl_operator_amd.ops.sampling import CompactionWorkspace
+
+    ws = CompactionWorkspace(1000, dev)
+    seeds = torch.tensor([7, 8], device=dev)
+    nbrs = torch.tensor([5, 6, 9], device=dev).repeat(100_000)
+    srcdata, local = ws.relabel(seeds, nbrs)
+    assert srcdata.numel() == 5  # 2 seeds + 3 unique neighbors
+    assert torch.equal(srcdata[:2].cpu(), torch.tensor([7, 8]))
+    assert torch.equal(srcdata[local], nbrs)
+    assert int((ws.table != -1).sum()) == 0
