@@ -106,7 +106,7 @@ class DistGraph:
     @staticmethod
     def from_partition(json_path: str, part_id: int, device="cpu") -> "DistGraph":
         gpart, feats, spec = load_partition(json_path, part_id)
-        book = PartitionBook(spec.boundaries)
+        book = PartitionBook(spec.boundaries, device=device)
         lo, hi = book.owned_range(part_id)
         src = gpart["src_global"].to(device)
         dst = (gpart["dst_global"] - lo).to(device)

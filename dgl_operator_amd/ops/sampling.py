@@ -4,9 +4,10 @@ Replaces ``dgl.distributed.sample_neighbors`` + ``dgl.to_block``
 (/root/reference/examples/GraphSAGE_dist/code/train_dist.py:52-70) with an
 entirely on-device path: a HIP sampling kernel (csrc/sampling.hip) draws
 ``fanout`` in-neighbors per seed (without replacement, DGL's default), and
-compaction/relabeling runs as vectorized torch ops on the GPU — no sampler
-worker processes (the reference needed --num-samplers CPU processes; on
-MI355X the sampler is a kernel).
+compaction/relabeling is a fused atomic-claim kernel chain (sample_block:
+one host sync per hop; sample_block_capture: zero, for hipGraph capture) —
+no sampler worker processes (the reference needed --num-samplers CPU
+processes; on MI355X the sampler is a kernel).
 """
 from __future__ import annotations
 
@@ -188,11 +189,9 @@ def sample_block_capture(
 class NeighborSampler:
     """Multi-layer neighbor sampler over a local CSC graph.
 
-    fanouts are listed INNERMOST-FIRST like the reference's --fan_out 10,25
-    (layer 0 fanout 10, layer 1 fanout 25? No: DGL's MultiLayerNeighborSampler
-    lists per-layer fanouts from the input layer to the output layer; the
-    reference passes [10, 25] meaning the first GNN layer aggregates 10-sampled
-    blocks... practically: we sample with fanouts reversed from seeds).
+    ``fanouts`` lists per-layer fanouts input-layer-first, like the
+    reference's --fan_out 10,25 (DGL MultiLayerNeighborSampler order), so
+    sampling proceeds over ``reversed(fanouts)`` starting from the seeds.
     Produces blocks ordered input-layer-first, as model.forward expects.
     """
 

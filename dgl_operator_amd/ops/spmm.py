@@ -8,8 +8,10 @@ and reduce ∈ {sum, mean}. This is the op behind
 (/root/reference/examples/GraphSAGE/code/3_message_passing.py:113,263) and the
 aggregation inside dgl.nn.SAGEConv / GraphConv.
 
-GPU path: hand-written HIP kernel (csrc/spmm.hip) over the CSC structure;
-backward runs the SAME kernel over the transposed (CSR) structure. CPU path:
+GPU path: hand-written HIP kernels (csrc/gnn_ops.hip) over the CSC
+structure; backward is a scatter-atomic walk of the SAME CSC (sampled
+blocks; note: float-atomic order makes it run-to-run nondeterministic at
+the ulp level) or the cached transposed CSR (full graphs / CPU). CPU path:
 pure-PyTorch fp32 reference (index_add_) used for numerics tests.
 """
 from __future__ import annotations

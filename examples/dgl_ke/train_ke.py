@@ -39,6 +39,8 @@ def main():
     p.add_argument("--num-triples", type=int, default=5_000_000)
     p.add_argument("--save-path", default="")
     p.add_argument("--no-save-emb", action="store_true")
+    p.add_argument("--json", action="store_true",
+                   help="print a bench-style JSON line (triples/s) at the end")
     p.add_argument("--eval", action="store_true",
                    help="report raw MRR/MR/Hits@K on held-out triples")
     p.add_argument("--num-eval", type=int, default=500)
@@ -82,6 +84,26 @@ def main():
             tps = step * args.batch_size * ws / elapsed
             print(f"step {step} loss {loss:.4f} {tps:,.0f} triples/s",
                   flush=True)
+    if args.json and rank == 0:
+        elapsed = time.time() - t0
+        tps = args.max_step * args.batch_size * ws / elapsed
+        print(json.dumps({
+            "metric": "triples/sec (whole node) KGE",
+            "value": tps,
+            "unit": "triples/s",
+            "n_gpus": ws,
+            "steps": args.max_step,
+            "ms_per_step": elapsed / args.max_step * 1000.0,
+            "higher_is_better": True,
+            "dtype": "fp32",
+            "data": "synthetic KG (uniform triples), random-init embeddings",
+            "config": {
+                "model": f"{args.model_name} d{args.hidden_dim} g{args.gamma}",
+                "global_batch": args.batch_size * ws,
+                "neg_sample_size": args.neg_sample_size,
+                "parallelism": f"sharded-kvstore ep{ws}",
+            },
+        }), flush=True)
     if args.eval:
         from dgl_operator_amd.distributed.kge import evaluate_kge
 
