@@ -9,7 +9,7 @@ all-reduce, feature/embedding pulls (alltoallv) and barriers.
 from __future__ import annotations
 
 import os
-from typing import List, Optional, Sequence, Tuple
+from typing import Optional, Sequence, Tuple
 
 import torch
 import torch.distributed as dist

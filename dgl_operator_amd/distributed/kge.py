@@ -13,13 +13,11 @@ NewBidirectionalOneShotIterator (hotfix/sampler.py:823-875).
 """
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Tuple
 
 import torch
 
 from ..ops import get_score_func, kge_loss
-from ..ops.adagrad import sparse_adagrad_update
-from . import comm
 from .kvstore import ShardedEmbedding
 
 

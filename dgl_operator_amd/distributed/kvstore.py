@@ -12,8 +12,6 @@ sharded 8-way).
 """
 from __future__ import annotations
 
-from typing import Optional, Tuple
-
 import torch
 
 from ..ops.adagrad import sparse_adagrad_update

@@ -8,7 +8,7 @@ machine_id partitioning).
 """
 from __future__ import annotations
 
-from typing import List, Sequence, Tuple
+from typing import Sequence, Tuple
 
 import torch
 

@@ -6,8 +6,6 @@ examples/graph_classification/code/5_graph_classification.py:155-166).
 """
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 import torch.nn as nn
 import torch.nn.functional as F

@@ -9,9 +9,6 @@ setting, entity embeddings live in the sharded KVStore
 """
 from __future__ import annotations
 
-import math
-from typing import Optional, Tuple
-
 import torch
 import torch.nn as nn
 

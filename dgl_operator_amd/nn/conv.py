@@ -12,8 +12,7 @@ Block (dst nodes are the first ``num_dst`` src rows).
 """
 from __future__ import annotations
 
-import math
-from typing import Optional, Union
+from typing import Union
 
 import torch
 import torch.nn as nn

@@ -11,7 +11,6 @@ implementations:
 """
 from __future__ import annotations
 
-import copy
 import json
 import subprocess
 from dataclasses import dataclass, field

@@ -22,7 +22,7 @@ from __future__ import annotations
 
 import copy
 import time
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 from .api import (
     CONFIG_MOUNT,

@@ -8,7 +8,6 @@ raises, rather than silently falling back to eager PyTorch.
 from __future__ import annotations
 
 import importlib
-import os
 from typing import Optional
 
 _EXT = None

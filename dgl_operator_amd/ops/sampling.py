@@ -11,7 +11,7 @@ processes; on MI355X the sampler is a kernel).
 """
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Tuple
 
 import torch
 

@@ -15,7 +15,6 @@ import time
 from contextlib import contextmanager
 
 from . import launch as launch_mod
-from .fabric import get_fabric
 from .hostfile import parse_hostfile, revise_for_dglke
 
 

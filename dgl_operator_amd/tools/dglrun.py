@@ -13,7 +13,6 @@ from __future__ import annotations
 import argparse
 import os
 import subprocess
-import sys
 import time
 from contextlib import contextmanager
 

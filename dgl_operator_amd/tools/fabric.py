@@ -16,9 +16,8 @@ import os
 import shlex
 import shutil
 import subprocess
-import threading
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional, Sequence, Tuple
+from typing import Dict, Optional
 
 
 class Fabric:
