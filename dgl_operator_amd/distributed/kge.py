@@ -141,12 +141,18 @@ def evaluate_kge(
     batch_size: int = 128,
     hits: Tuple[int, ...] = (1, 3, 10),
     corrupt: str = "tail",
+    filter_triples: "Tuple[torch.Tensor, torch.Tensor, torch.Tensor] | None" = None,
 ):
-    """Raw-setting link-prediction metrics (MRR, MR, Hits@K) — the dglke eval
-    protocol (reference EvalSampler/EvalDataset, hotfix/sampler.py:514-821):
-    rank the true entity among ALL entities. With sharded entity tables each
-    rank scores the candidates it OWNS and the global rank is the all-reduced
+    """Link-prediction metrics (MRR, MR, Hits@K) — the dglke eval protocol
+    (reference EvalSampler/EvalDataset, hotfix/sampler.py:514-821): rank the
+    true entity among ALL entities. With sharded entity tables each rank
+    scores the candidates it OWNS and the global rank is the all-reduced
     count of higher-scoring candidates + 1.
+
+    ``filter_triples`` switches to the FILTERED setting: known triples
+    (h, r, t') with t' != t are excluded from the ranking by subtracting the
+    known competitors that outscore the true entity (every rank is assumed
+    to hold the full triple list, as in this repo's synthetic KGs).
     """
     import torch.distributed as dist
 
@@ -157,6 +163,18 @@ def evaluate_kge(
     n_total = heads.numel()
     shard = model.entities.local  # [M, D] owned candidate entities
     M = shard.shape[0]
+    known = None
+    if filter_triples is not None:
+        from collections import defaultdict
+
+        kh, kr, kt = filter_triples
+        known = defaultdict(list)
+        if corrupt == "tail":
+            for a, b, c in zip(kh.tolist(), kr.tolist(), kt.tolist()):
+                known[(a, b)].append(c)
+        else:
+            for a, b, c in zip(kh.tolist(), kr.tolist(), kt.tolist()):
+                known[(c, b)].append(a)
     for s in range(0, n_total, batch_size):
         hh = heads[s : s + batch_size]
         rr = rels[s : s + batch_size]
@@ -179,6 +197,27 @@ def evaluate_kge(
         higher = (cand > true_score.unsqueeze(1)).sum(1).double()
         if dist.is_available() and dist.is_initialized():
             dist.all_reduce(higher)
+        if known is not None:
+            # filtered: remove known competitors that outscored the truth
+            adj = torch.zeros_like(higher)
+            for i in range(B):
+                key = (int(hh[i]), int(rr[i]))
+                cand_ids = [e for e in known.get(key, ())
+                            if e != int(tt[i] if corrupt == "tail" else hh[i])]
+                if not cand_ids:
+                    continue
+                ids_t = torch.as_tensor(cand_ids, device=device)
+                ke = model.entities.pull(ids_t)
+                if corrupt == "tail":
+                    sc = model.score.edge(
+                        h[i].expand(len(cand_ids), -1),
+                        r[i].expand(len(cand_ids), -1), ke)
+                else:
+                    sc = model.score.edge(
+                        ke, r[i].expand(len(cand_ids), -1),
+                        t[i].expand(len(cand_ids), -1))
+                adj[i] = (sc > true_score[i]).sum()
+            higher = (higher - adj).clamp(min=0)
         ranks = higher + 1.0
         rank_sum += float(ranks.sum())
         rr_sum += float((1.0 / ranks).sum())

@@ -242,7 +242,12 @@ class DGLJobReconciler:
         spec_t = job.spec.replica_specs.get(ReplicaType.LAUNCHER)
         template = copy.deepcopy(spec_t.template) if spec_t else {}
         pod_spec = template.get("spec", {})
-        init_containers = [{"name": "kubectl-download", "image": "kubectl-download"}]
+        init_resources = {  # fixed init-container resources
+            "cpu": "100m", "memory": "512Mi", "ephemeral-storage": "5Gi",
+        }  # dgljob_controller.go:74-76,1139-1150
+        init_containers = [{"name": "kubectl-download",
+                            "image": "kubectl-download",
+                            "resources": init_resources}]
         if job.spec.partition_mode in (PartitionMode.DGL_API, PartitionMode.PARMETIS):
             # watcher-loop-partitioner also mounts the dataset volume so the
             # partitioner can copy partitions into this still-running init
@@ -252,12 +257,19 @@ class DGLJobReconciler:
                 "image": "watcher-loop",
                 "env": {"WATCHERFILE": "partfile", "WATCHERMODE": "finished"},
                 "mounts": ["dataset"],
+                "resources": init_resources,
             })
         init_containers.append({
             "name": "watcher-loop-worker",
             "image": "watcher-loop",
             "env": {"WATCHERFILE": "hostfile", "WATCHERMODE": "ready"},
+            "resources": init_resources,
         })
+        # launcher main container defaults to 1 CPU / 2Gi when unset
+        # (dgljob_controller.go:77-78,1229-1240)
+        for cont in pod_spec.get("containers", []):
+            cont.setdefault("resources",
+                            {"limits": {"cpu": "1", "memory": "2Gi"}})
         env = dict(pod_spec.get("env", {}))
         env[ENV_KUBEXEC_PATH] = f"{CONFIG_MOUNT}/kubexec.sh"
         env[ENV_KUBECTL_PATH] = f"{KUBECTL_MOUNT}/kubectl"
