@@ -17,6 +17,8 @@ from typing import Union
 import torch
 import torch.nn as nn
 
+import torch.nn.functional as F
+
 from ..graph.graph import Block, Graph
 from ..ops import gspmm, edge_softmax_csc
 
@@ -53,12 +55,24 @@ class SAGEConv(nn.Module):
                 edge_weight: torch.Tensor = None) -> torch.Tensor:
         """Optional per-edge scalar weight gives the reference's
         WeightedSAGEConv (u_mul_e aggregate,
-        /root/reference/examples/GraphSAGE/code/3_message_passing.py:263)."""
-        if edge_weight is None:
-            h_neigh = gspmm(g, "copy_u", self.aggregator, x)
+        /root/reference/examples/GraphSAGE/code/3_message_passing.py:263).
+
+        When in_feats > out_feats the projection runs BEFORE aggregation:
+        sum/mean are linear, so mean(x W) + b == fc_neigh(mean(x)), and the
+        SpMM moves out_feats-wide rows instead of in_feats-wide ones (6x
+        less HBM traffic at the bench shape 100 -> 16). Zero-degree rows
+        match too: both orders yield exactly the bias."""
+        op = "copy_u" if edge_weight is None else "u_mul_e"
+        if self.fc_neigh.in_features > self.fc_neigh.out_features:
+            pre = F.linear(x, self.fc_neigh.weight)  # bias added after
+            h_neigh = gspmm(g, op, self.aggregator, pre, edge_weight)
+            if self.fc_neigh.bias is not None:
+                h_neigh = h_neigh + self.fc_neigh.bias
         else:
-            h_neigh = gspmm(g, "u_mul_e", self.aggregator, x, edge_weight)
-        return self.fc_self(_dst_feat(g, x)) + self.fc_neigh(h_neigh)
+            h_neigh = self.fc_neigh(
+                gspmm(g, op, self.aggregator, x, edge_weight)
+            )
+        return self.fc_self(_dst_feat(g, x)) + h_neigh
 
 
 class GraphConv(nn.Module):

@@ -118,3 +118,27 @@ def test_weighted_sageconv():
     assert out.shape == (50, 4)
     out.sum().backward()
     assert w.grad is not None
+
+
+def test_sageconv_projection_order_equivalence():
+    """project-then-aggregate == aggregate-then-project (mean is linear),
+    including zero-degree destinations (bias-only rows)."""
+    from dgl_operator_amd.nn import SAGEConv
+    from dgl_operator_amd.ops import gspmm
+
+    torch.manual_seed(4)
+    g = rmat_graph(60, 300, seed=6)  # has zero-in-degree nodes
+    assert (g.in_degrees() == 0).any()
+    layer = SAGEConv(20, 4)  # in > out: fast path active
+    x = torch.randn(60, 20)
+    out_fast = layer(g, x)
+    # reference order: aggregate 20-dim, then project
+    h_n = gspmm(g, "copy_u", "mean", x)
+    out_ref = layer.fc_self(x) + layer.fc_neigh(h_n)
+    assert torch.allclose(out_fast, out_ref, atol=1e-5)
+    # weighted variant
+    w = torch.rand(g.num_edges)
+    out_fast_w = layer(g, x, edge_weight=w)
+    h_nw = gspmm(g, "u_mul_e", "mean", x, w)
+    out_ref_w = layer.fc_self(x) + layer.fc_neigh(h_nw)
+    assert torch.allclose(out_fast_w, out_ref_w, atol=1e-5)
