@@ -14,6 +14,8 @@ at::Tensor edge_softmax_fwd(at::Tensor indptr, at::Tensor scores);
 at::Tensor edge_softmax_bwd(at::Tensor indptr, at::Tensor out,
                             at::Tensor grad_out);
 at::Tensor segment_reduce(at::Tensor offsets, at::Tensor feat, bool mean);
+at::Tensor gather_rows(at::Tensor feat, at::Tensor gids,
+                       c10::optional<at::Tensor> map, int64_t offset);
 std::tuple<at::Tensor, at::Tensor> sample_neighbors(at::Tensor indptr,
                                                     at::Tensor indices,
                                                     at::Tensor seeds,
@@ -60,6 +62,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("edge_softmax_fwd", &doa::edge_softmax_fwd);
   m.def("edge_softmax_bwd", &doa::edge_softmax_bwd);
   m.def("segment_reduce", &doa::segment_reduce);
+  m.def("gather_rows", &doa::gather_rows, py::arg("feat"), py::arg("gids"),
+        py::arg("map") = py::none(), py::arg("offset") = 0);
   m.def("sample_neighbors", &doa::sample_neighbors);
   m.def("compact_ids", &doa::compact_ids);
   m.def("sample_block", &doa::sample_block, py::arg("indptr"),

@@ -557,3 +557,72 @@ at::Tensor segment_reduce(at::Tensor offsets, at::Tensor feat, bool mean) {
 }
 
 }  // namespace doa
+
+namespace doa {
+
+// ---------------------------------------------------------------------------
+// Fused row gather: out[i, :] = feat[(map ? map[gids[i]] : gids[i] - offset), :]
+// One kernel replaces the two-gather chain of the halo-cache feature pull
+// (feat[feat_map[gids]]) and the sub+index_select of the owned-range pull.
+// ---------------------------------------------------------------------------
+template <typename scalar_t, int VEC>
+__global__ void gather_rows_kernel(const scalar_t* __restrict__ feat,
+                                   const int64_t* __restrict__ gids,
+                                   const int64_t* __restrict__ map,
+                                   scalar_t* __restrict__ out, int64_t n,
+                                   int F, int64_t offset) {
+  const int chunks = F / VEC;
+  const int64_t total = n * chunks;
+  for (int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       tid < total; tid += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t i = tid / chunks;
+    const int f0 = (int)(tid % chunks) * VEC;
+    const int64_t g = gids[i];
+    const int64_t row = map ? map[g] : (g - offset);
+    const scalar_t* src = feat + row * F + f0;
+    scalar_t* dst = out + i * F + f0;
+#pragma unroll
+    for (int k = 0; k < VEC; ++k) dst[k] = src[k];
+  }
+}
+
+at::Tensor gather_rows(at::Tensor feat, at::Tensor gids,
+                       c10::optional<at::Tensor> map, int64_t offset) {
+  TORCH_CHECK(feat.is_cuda(), "gather_rows: feat must be on GPU");
+  auto f = feat.contiguous();
+  const int64_t n = gids.numel();
+  int F = 1;
+  for (int i = 1; i < f.dim(); ++i) F *= f.size(i);
+  std::vector<int64_t> osz;
+  osz.push_back(n);
+  for (int i = 1; i < f.dim(); ++i) osz.push_back(f.size(i));
+  auto out = at::empty(osz, f.options());
+  const int block = 256;
+  auto stream = cur_stream();
+  const int64_t* mp = map.has_value() ? map->data_ptr<int64_t>() : nullptr;
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half, f.scalar_type(), "gather_rows", [&] {
+    if (F % 4 == 0 && sizeof(scalar_t) == 4) {
+      hipLaunchKernelGGL((gather_rows_kernel<scalar_t, 4>),
+                         dim3(grid_for(n * (F / 4), block)), dim3(block), 0,
+                         stream, f.data_ptr<scalar_t>(),
+                         gids.data_ptr<int64_t>(), mp,
+                         out.data_ptr<scalar_t>(), n, F, offset);
+    } else if (F % 8 == 0 && sizeof(scalar_t) == 2) {
+      hipLaunchKernelGGL((gather_rows_kernel<scalar_t, 8>),
+                         dim3(grid_for(n * (F / 8), block)), dim3(block), 0,
+                         stream, f.data_ptr<scalar_t>(),
+                         gids.data_ptr<int64_t>(), mp,
+                         out.data_ptr<scalar_t>(), n, F, offset);
+    } else {
+      hipLaunchKernelGGL((gather_rows_kernel<scalar_t, 1>),
+                         dim3(grid_for(n * F, block)), dim3(block), 0,
+                         stream, f.data_ptr<scalar_t>(),
+                         gids.data_ptr<int64_t>(), mp,
+                         out.data_ptr<scalar_t>(), n, F, offset);
+    }
+  });
+  DOA_CHECK_HIP(hipGetLastError());
+  return out;
+}
+
+}  // namespace doa

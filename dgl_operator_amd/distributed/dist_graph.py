@@ -219,9 +219,20 @@ class DistGraph:
         halo-cached keys gather locally instead."""
         rank, ws = comm.world()
         if self.halo is not None and key in self.halo.feats:
-            return self.halo.feats[key][self.halo.feat_map[gids]]
+            cached = self.halo.feats[key]
+            if gids.is_cuda and cached.is_floating_point():
+                from ..ops import backend
+
+                ext = backend.ext_for(gids)
+                return ext.gather_rows(cached, gids, self.halo.feat_map, 0)
+            return cached[self.halo.feat_map[gids]]
         feat = self.ndata[key]
         if ws == 1:
+            if gids.is_cuda and feat.is_floating_point():
+                from ..ops import backend
+
+                ext = backend.ext_for(gids)
+                return ext.gather_rows(feat, gids, None, self.lo)
             return feat[gids - self.lo]
         sorted_ids, perm, send_counts = self.book.partition_by_owner(gids)
         recv_counts = comm.exchange_counts(send_counts)

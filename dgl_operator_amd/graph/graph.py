@@ -57,6 +57,7 @@ class Graph:
         self._csc: Optional[Tuple[torch.Tensor, torch.Tensor, torch.Tensor]] = None
         # CSR: per-src out-neighbors (backward / transposed structure)
         self._csr: Optional[Tuple[torch.Tensor, torch.Tensor, torch.Tensor]] = None
+        self._csc_dst: Optional[torch.Tensor] = None
         self.ndata = {}
         self.edata = {}
 
@@ -89,6 +90,16 @@ class Graph:
         if self._csr is None:
             self._csr = _coo_to_compressed(self._src, self._dst, self._num_nodes)
         return self._csr
+
+    def csc_dst(self) -> torch.Tensor:
+        """Cached per-CSC-position destination node id."""
+        if self._csc_dst is None:
+            indptr, _, _ = self.csc()
+            self._csc_dst = torch.repeat_interleave(
+                torch.arange(indptr.numel() - 1, device=indptr.device),
+                indptr[1:] - indptr[:-1],
+            )
+        return self._csc_dst
 
     def in_degrees(self) -> torch.Tensor:
         indptr, _, _ = self.csc()
@@ -201,6 +212,7 @@ class Block:
         self.srcdata_nids = srcdata_nids
         self.csc_eids = eids
         self._csr: Optional[Tuple[torch.Tensor, torch.Tensor, torch.Tensor]] = None
+        self._csc_dst: Optional[torch.Tensor] = None
         self.edata = {}
 
     @property
@@ -232,6 +244,15 @@ class Block:
             )
             self._csr = _coo_to_compressed(self.csc_indices, dst, self._num_src)
         return self._csr
+
+    def csc_dst(self) -> torch.Tensor:
+        """Cached per-CSC-position destination (block-local) node id."""
+        if self._csc_dst is None:
+            self._csc_dst = torch.repeat_interleave(
+                torch.arange(self._num_dst, device=self.device),
+                self.csc_indptr[1:] - self.csc_indptr[:-1],
+            )
+        return self._csc_dst
 
     def in_degrees(self) -> torch.Tensor:
         return self.csc_indptr[1:] - self.csc_indptr[:-1]

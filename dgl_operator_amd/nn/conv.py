@@ -149,10 +149,8 @@ class GATConv(nn.Module):
         z = self.fc(x).view(-1, H, D)  # [N, H, D]
         el = (z * self.attn_l).sum(-1)  # [N, H]
         er = (z * self.attn_r).sum(-1)
-        indptr, indices, _ = g.csc()
-        from ..ops.spmm import _edge_dst
-
-        dst = _edge_dst(indptr)
+        _, indices, _ = g.csc()
+        dst = g.csc_dst()
         score = self.leaky(el[indices] + er[dst])  # [E, H], csc order
         alpha = edge_softmax_csc(g, score)
         out = gspmm(g, "u_mul_e", "sum", z, _csc_weight(g, alpha))  # [Nd, H, D]

@@ -351,3 +351,24 @@ def test_compact_ids_heavy_duplicates(dev):
     assert torch.equal(srcdata[:2].cpu(), torch.tensor([7, 8]))
     assert torch.equal(srcdata[local], nbrs)
     assert int((ws.table != -1).sum()) == 0
+
+
+def test_gather_rows_kernel(dev):
+    ext = backend.load_extension(required=True)
+    feat = torch.randn(1000, 100, device=dev)
+    gids = torch.randint(50, 1050, (3000,), device=dev)
+    # offset form
+    out = ext.gather_rows(feat, gids, None, 50)
+    assert torch.equal(out, feat[gids - 50])
+    # map form
+    mp = torch.randperm(2000, device=dev)[:1050]
+    mp_full = torch.full((2000,), -1, dtype=torch.int64, device=dev)
+    mp_full[:1050] = torch.randint(0, 1000, (1050,), device=dev)
+    out2 = ext.gather_rows(feat, gids.clamp(max=1049), mp_full, 0)
+    assert torch.equal(out2, feat[mp_full[gids.clamp(max=1049)]])
+    # odd width (VEC=1 path) + bf16 (VEC=8)
+    f2 = torch.randn(100, 7, device=dev)
+    ids = torch.arange(100, device=dev)
+    assert torch.equal(ext.gather_rows(f2, ids, None, 0), f2)
+    f3 = torch.randn(100, 64, device=dev).bfloat16()
+    assert torch.equal(ext.gather_rows(f3, ids, None, 0), f3)
