@@ -372,3 +372,25 @@ def test_gather_rows_kernel(dev):
     assert torch.equal(ext.gather_rows(f2, ids, None, 0), f2)
     f3 = torch.randn(100, 64, device=dev).bfloat16()
     assert torch.equal(ext.gather_rows(f3, ids, None, 0), f3)
+
+
+def test_bench_capture_mode_gpu(dev):
+    """--capture (whole-step hipGraph replay) produces a sane bench line."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--capture", "--steps", "4",
+         "--warmup", "2", "--nodes", "20000", "--edges", "150000"],
+        capture_output=True, text=True, cwd=repo, timeout=600,
+    )
+    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+    d = json.loads([l for l in r.stdout.splitlines() if l.startswith("{")][-1])
+    assert d["value"] > 0
+    # edges per step must be plausible: ~batch * (25 + 10*frontier)-ish, and
+    # device-side accounting must not count the padded garbage region
+    edges_per_step = d["value"] * d["ms_per_step"] / 1000.0
+    assert 1000 < edges_per_step < 1000 * 36
