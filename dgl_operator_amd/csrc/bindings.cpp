@@ -16,6 +16,8 @@ at::Tensor edge_softmax_bwd(at::Tensor indptr, at::Tensor out,
 at::Tensor segment_reduce(at::Tensor offsets, at::Tensor feat, bool mean);
 at::Tensor gather_rows(at::Tensor feat, at::Tensor gids,
                        c10::optional<at::Tensor> map, int64_t offset);
+at::Tensor gather_mm(at::Tensor feat, at::Tensor rows, at::Tensor weight,
+                     c10::optional<at::Tensor> bias);
 std::tuple<at::Tensor, at::Tensor> sample_neighbors(at::Tensor indptr,
                                                     at::Tensor indices,
                                                     at::Tensor seeds,
@@ -64,6 +66,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("segment_reduce", &doa::segment_reduce);
   m.def("gather_rows", &doa::gather_rows, py::arg("feat"), py::arg("gids"),
         py::arg("map") = py::none(), py::arg("offset") = 0);
+  m.def("gather_mm", &doa::gather_mm, py::arg("feat"), py::arg("rows"),
+        py::arg("weight"), py::arg("bias") = py::none());
   m.def("sample_neighbors", &doa::sample_neighbors);
   m.def("compact_ids", &doa::compact_ids);
   m.def("sample_block", &doa::sample_block, py::arg("indptr"),

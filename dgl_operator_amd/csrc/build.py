@@ -24,6 +24,7 @@ SOURCES = [
     "sampling.hip",
     "adagrad.hip",
     "kge.hip",
+    "gather_mm.hip",
     "ldg.cpp",
     "bindings.cpp",
 ]

@@ -243,6 +243,20 @@ class DistGraph:
         out[perm] = rows_back
         return out
 
+    def pull_view(self, key: str, gids: torch.Tensor):
+        """Lazy GatherView over the halo cache (or the owned shard when
+        world==1): lets fusion-aware layers project straight from the table
+        (ops.gather_mm) without materializing the gathered features. Falls
+        back to an eager pull when the ids are not locally resolvable."""
+        from ..ops.gather_mm import GatherView
+
+        rank, ws = comm.world()
+        if self.halo is not None and key in self.halo.feats:
+            return GatherView(self.halo.feats[key], self.halo.feat_map[gids])
+        if ws == 1:
+            return GatherView(self.ndata[key], gids - self.lo)
+        return self.pull(key, gids)
+
     # -- distributed push (KVStore PUSH, C4) -------------------------------
     def push_accumulate(self, key: str, gids: torch.Tensor, rows: torch.Tensor):
         """Scatter-add rows into the owners' feature shards."""

@@ -21,6 +21,7 @@ import torch.nn.functional as F
 
 from ..graph.graph import Block, Graph
 from ..ops import gspmm, edge_softmax_csc
+from ..ops.gather_mm import GatherView
 
 
 def _num_dst(g) -> int:
@@ -63,6 +64,24 @@ class SAGEConv(nn.Module):
         less HBM traffic at the bench shape 100 -> 16). Zero-degree rows
         match too: both orders yield exactly the bias."""
         op = "copy_u" if edge_weight is None else "u_mul_e"
+        if isinstance(x, GatherView):
+            fusable = (
+                x.feat.is_cuda
+                and x.feat.dtype == torch.float32
+                and self.fc_neigh.out_features <= 16
+            )
+            if not fusable:
+                x = x.materialize()
+        if isinstance(x, GatherView):
+            # fused input projections straight from the feature table (MFMA
+            # gather_mm): no materialized x, projection before aggregation
+            pre = x.project(self.fc_neigh.weight.t())
+            h_neigh = gspmm(g, op, self.aggregator, pre, edge_weight)
+            if self.fc_neigh.bias is not None:
+                h_neigh = h_neigh + self.fc_neigh.bias
+            nd = g.num_dst_nodes if isinstance(g, Block) else g.num_nodes
+            h_self = x.narrow_rows(nd).project(self.fc_self.weight.t())
+            return h_self + h_neigh
         if self.fc_neigh.in_features > self.fc_neigh.out_features:
             pre = F.linear(x, self.fc_neigh.weight)  # bias added after
             h_neigh = gspmm(g, op, self.aggregator, pre, edge_weight)

@@ -11,6 +11,7 @@ from .sampling import (
 )
 from .kge import get_score_func, kge_loss, SCORE_FUNCS
 from .adagrad import sparse_adagrad_update
+from .gather_mm import gather_mm, GatherView
 
 __all__ = [
     "backend",
@@ -30,4 +31,6 @@ __all__ = [
     "kge_loss",
     "SCORE_FUNCS",
     "sparse_adagrad_update",
+    "gather_mm",
+    "GatherView",
 ]

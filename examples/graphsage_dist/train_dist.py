@@ -120,7 +120,7 @@ def main():
             inp, out_nodes, blocks = dg.sample_blocks(
                 seeds, fanouts, seed=epoch * 100000 + step
             )
-            x = dg.pull("feat", inp)
+            x = dg.pull_view("feat", inp)
             y = dg.pull("label", out_nodes)
             t_sample = time.time()
             logits = model(blocks, x)
