@@ -70,6 +70,15 @@ class Graph:
     def num_edges(self) -> int:
         return self._src.numel()
 
+    # homogeneous graphs are their own src/dst sets (DGL API uniformity)
+    @property
+    def num_src_nodes(self) -> int:
+        return self._num_nodes
+
+    @property
+    def num_dst_nodes(self) -> int:
+        return self._num_nodes
+
     @property
     def device(self) -> torch.device:
         return self._src.device
@@ -150,20 +159,47 @@ def _update_all(g, message_func, reduce_func, num_dst: int) -> None:
     from ..ops import gspmm
     from ..ops.udf import update_all_udf
 
+    is_block = isinstance(g, Block)
+    src_fields = g.srcdata if is_block else g.ndata
+    out_fields = g.dstdata if is_block else g.ndata
     if isinstance(message_func, fnmod.MessageFn) and isinstance(
         reduce_func, fnmod.ReduceFn
     ):
         assert message_func.out_field == reduce_func.msg_field, (
             "message out field must feed the reduce"
         )
-        feat = g.ndata[message_func.src_field]
+        feat = src_fields[message_func.src_field]
         w = g.edata[message_func.edge_field] if message_func.op == "u_mul_e" else None
-        g.ndata[reduce_func.out_field] = gspmm(
+        out_fields[reduce_func.out_field] = gspmm(
             g, message_func.op, reduce_func.op, feat, w
         )
         return
-    out = update_all_udf(g, g.ndata, g.edata, message_func, reduce_func, num_dst)
-    g.ndata.update(out)
+    out = update_all_udf(g, src_fields, g.edata, message_func, reduce_func,
+                         num_dst)
+    out_fields.update(out)
+
+
+def _apply_edges_block(b, edge_func) -> None:
+    from . import graph as _self  # noqa: F401
+    from .. import fn as fnmod
+    from ..ops import sddmm_dot
+    from ..ops.udf import EdgeBatch
+
+    if isinstance(edge_func, fnmod.EdgeFn):
+        assert edge_func.op == "u_dot_v"
+        b.edata[edge_func.out_field] = sddmm_dot(
+            b, b.srcdata[edge_func.lhs_field], b.dstdata[edge_func.rhs_field]
+        )
+        return
+    src = b.csc_indices
+    dst = b.csc_dst()
+    batch = EdgeBatch(
+        {k: v[src] for k, v in b.srcdata.items()},
+        {k: v[dst] for k, v in b.dstdata.items()},
+        dict(b.edata),
+    )
+    for k, v in edge_func(batch).items():
+        b.edata[k] = v
 
 
 def _apply_edges(g, edge_func) -> None:
@@ -214,6 +250,8 @@ class Block:
         self._csr: Optional[Tuple[torch.Tensor, torch.Tensor, torch.Tensor]] = None
         self._csc_dst: Optional[torch.Tensor] = None
         self.edata = {}
+        self.srcdata = {}  # per-src-node fields (block-local rows)
+        self.dstdata = {}  # per-dst-node fields
 
     @property
     def num_src_nodes(self) -> int:
@@ -256,6 +294,14 @@ class Block:
 
     def in_degrees(self) -> torch.Tensor:
         return self.csc_indptr[1:] - self.csc_indptr[:-1]
+
+    def update_all(self, message_func, reduce_func) -> None:
+        """Blocks support the fn API too; reduced fields land in
+        ``self.dstdata`` (dst rows only)."""
+        _update_all(self, message_func, reduce_func, self._num_dst)
+
+    def apply_edges(self, edge_func) -> None:
+        _apply_edges_block(self, edge_func)
 
     def to(self, device) -> "Block":
         b = Block(

@@ -75,3 +75,23 @@ def test_udf_autograd_flows():
 
 def test_copy_src_alias():
     assert fn.copy_src("h", "m") == fn.copy_u("h", "m")
+
+
+def test_block_fn_api():
+    from dgl_operator_amd.ops import NeighborSampler
+
+    g = rmat_graph(50, 400, num_feats=6, seed=3)
+    indptr, indices, _ = g.csc()
+    sampler = NeighborSampler(indptr, indices, [4], num_nodes=50)
+    inp, out, (blk,) = sampler.sample_blocks(torch.arange(20))
+    blk.srcdata["h"] = g.ndata["feat"][blk.srcdata_nids]
+    blk.dstdata["h"] = blk.srcdata["h"][: blk.num_dst_nodes]
+    blk.update_all(fn.copy_u("h", "m"), fn.mean("m", "h_N"))
+    assert blk.dstdata["h_N"].shape == (20, 6)
+    # matches direct gspmm
+    from dgl_operator_amd.ops import gspmm
+
+    ref = gspmm(blk, "copy_u", "mean", blk.srcdata["h"])
+    assert torch.allclose(blk.dstdata["h_N"], ref, atol=1e-6)
+    blk.apply_edges(fn.u_dot_v("h", "h", "score"))
+    assert blk.edata["score"].shape == (blk.num_edges,)
