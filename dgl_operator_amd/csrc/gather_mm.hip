@@ -107,7 +107,9 @@ at::Tensor gather_mm(at::Tensor feat, at::Tensor rows, at::Tensor weight,
   const int Kp = (K + 3) & ~3;
   const size_t lds_bytes =
       (GM_ROWS * (Kp + 1) + Kp * GM_N) * sizeof(float);
-  TORCH_CHECK(lds_bytes <= 160 * 1024, "gather_mm: K too large for LDS");
+  // HIP caps dynamic LDS at 64 KiB unless the max-dynamic-shared attribute
+  // is raised; K <= ~230 fits comfortably under the default cap
+  TORCH_CHECK(lds_bytes <= 64 * 1024, "gather_mm: K too large for LDS");
   const int grid = grid_for(ceil_div(M, GM_ROWS) * 256, 256);
   auto stream = c10::hip::getCurrentHIPStream().stream();
   at::Tensor bc;  // keep the contiguous bias alive across the async launch
