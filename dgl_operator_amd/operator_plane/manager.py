@@ -56,9 +56,13 @@ class _HealthHandler(BaseHTTPRequestHandler):
 
 class Manager:
     def __init__(self, cluster: Optional[Cluster] = None,
-                 reconcile_interval: float = 0.5):
+                 reconcile_interval: float = 0.5,
+                 watcher_loop_image: str = "watcher-loop",
+                 kubectl_download_image: str = "kubectl-download"):
         self.cluster = cluster or FakeCluster()
-        self.reconciler = DGLJobReconciler(self.cluster)
+        self.reconciler = DGLJobReconciler(
+            self.cluster, watcher_loop_image, kubectl_download_image
+        )
         self.jobs: Dict[str, DGLJob] = {}
         self.interval = reconcile_interval
         self.ready = False
@@ -128,10 +132,17 @@ def main(argv=None):
     p.add_argument("--metrics-bind-address", default=":8080")
     p.add_argument("--health-probe-bind-address", default=":8081")
     p.add_argument("--reconcile-interval", type=float, default=0.5)
+    p.add_argument("--leader-elect", action="store_true",
+                   help="accepted for flag parity with the reference manager; "
+                        "single-replica deployments need no election")
+    p.add_argument("--watcher-loop-image", default="watcher-loop")
+    p.add_argument("--kubectl-download-image", default="kubectl-download")
     p.add_argument("--job", action="append", default=[],
                    help="DGLJob manifest YAML file(s) to manage")
     args = p.parse_args(argv)
-    mgr = Manager(reconcile_interval=args.reconcile_interval)
+    mgr = Manager(reconcile_interval=args.reconcile_interval,
+                  watcher_loop_image=args.watcher_loop_image,
+                  kubectl_download_image=args.kubectl_download_image)
     for path in args.job:
         with open(path) as f:
             mgr.submit(f.read())

@@ -57,8 +57,14 @@ shift
 
 
 class DGLJobReconciler:
-    def __init__(self, cluster: Cluster):
+    def __init__(self, cluster: Cluster,
+                 watcher_loop_image: str = "watcher-loop",
+                 kubectl_download_image: str = "kubectl-download"):
         self.cluster = cluster
+        # operator flags --watcher-loop-image / --kubectl-download-image
+        # (main.go:52-69)
+        self.watcher_loop_image = watcher_loop_image
+        self.kubectl_download_image = kubectl_download_image
 
     # ------------------------------------------------------------------
     def reconcile(self, job: DGLJob) -> DGLJob:
@@ -246,7 +252,7 @@ class DGLJobReconciler:
             "cpu": "100m", "memory": "512Mi", "ephemeral-storage": "5Gi",
         }  # dgljob_controller.go:74-76,1139-1150
         init_containers = [{"name": "kubectl-download",
-                            "image": "kubectl-download",
+                            "image": self.kubectl_download_image,
                             "resources": init_resources}]
         if job.spec.partition_mode in (PartitionMode.DGL_API, PartitionMode.PARMETIS):
             # watcher-loop-partitioner also mounts the dataset volume so the
@@ -254,14 +260,14 @@ class DGLJobReconciler:
             # container (dgljob_controller.go:1129-1138)
             init_containers.append({
                 "name": "watcher-loop-partitioner",
-                "image": "watcher-loop",
+                "image": self.watcher_loop_image,
                 "env": {"WATCHERFILE": "partfile", "WATCHERMODE": "finished"},
                 "mounts": ["dataset"],
                 "resources": init_resources,
             })
         init_containers.append({
             "name": "watcher-loop-worker",
-            "image": "watcher-loop",
+            "image": self.watcher_loop_image,
             "env": {"WATCHERFILE": "hostfile", "WATCHERMODE": "ready"},
             "resources": init_resources,
         })
