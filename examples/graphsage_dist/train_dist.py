@@ -151,6 +151,23 @@ def main():
                     f"fwd {t_fwd - t_sample:.3f}s bwd+upd {t_bwd - t_fwd:.3f}s",
                     flush=True,
                 )
+        if args.eval_every and (epoch + 1) % args.eval_every == 0:
+            from dgl_operator_amd.models.graphsage import inference_dist
+
+            logits_shard = inference_dist(model, dg, batch_size=4096)
+            pred = logits_shard.argmax(1)
+            labels_local = dg.ndata["label"]
+            correct = (pred == labels_local).sum()
+            total = torch.tensor(
+                [float(correct), float(labels_local.numel())], device=device
+            )
+            if ws > 1:
+                dist.all_reduce(total)
+            if rank == 0:
+                print(f"Epoch {epoch:03d} | Eval acc "
+                      f"{float(total[0]) / max(float(total[1]), 1):.4f}",
+                      flush=True)
+            model.train()
         if rank == 0:
             print(f"Epoch {epoch:03d} time {time.time() - t_epoch:.2f}s",
                   flush=True)
