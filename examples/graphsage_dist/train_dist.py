@@ -60,10 +60,16 @@ def main():
     else:
         device = torch.device("cpu")
 
-    # partition id = node rank (one partition per worker pod, slots GPUs each;
-    # with slots=1 this is rank)
-    part_id = int(os.environ.get("GROUP_RANK", os.environ.get("RANK", rank)))
+    # one partition per RANK: the partition book maps global rank -> owned
+    # range, so the job must be partitioned into world_size parts
+    # (slotsPerWorker=1 deployments: rank == worker pod index)
+    part_id = rank
     dg = DistGraph.from_partition(args.part_config, part_id, device=device)
+    if ws > 1:
+        assert dg.book.num_parts == ws, (
+            f"partition count {dg.book.num_parts} != world size {ws}; "
+            "run the partitioner with --num-partitions == total ranks"
+        )
     if ws > 1 and not args.no_halo:
         dg.build_halo_cache(args.num_layers, feat_keys=("feat", "label"))
 
