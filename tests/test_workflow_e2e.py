@@ -79,3 +79,27 @@ def test_dglrun_launcher_end_to_end(tmp_path):
         pd = pods_root / f"job-worker-{i}"
         assert (pd / workspace / "workload" / f"part{i}" / "graph.pt").exists()
         assert (pd / workspace / "hostfile_revised").exists()
+
+
+@pytest.mark.timeout(300)
+def test_train_dist_single_node_multi_rank(tmp_path):
+    """torchrun --nproc-per-node 2 on one node: rank->partition mapping +
+    halo build + eval path (regression: GROUP_RANK used to alias part 0)."""
+    r = subprocess.run(
+        [sys.executable,
+         os.path.join(REPO, "examples/graphsage_dist/load_and_partition_graph.py"),
+         "--graph-name", "t", "--num-partitions", "2",
+         "--output", str(tmp_path), "--nodes", "300", "--edges", "2000",
+         "--feat", "8", "--classes", "3", "--algorithm", "range"],
+        capture_output=True, text=True, cwd=REPO)
+    assert r.returncode == 0, r.stderr
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--standalone",
+         "--local-addr", "127.0.0.1", "--nproc-per-node", "2",
+         os.path.join(REPO, "examples/graphsage_dist/train_dist.py"),
+         "--part-config", str(tmp_path / "t.json"),
+         "--num-epochs", "1", "--batch-size", "32", "--fan-out", "3,3",
+         "--eval-every", "1", "--log-every", "100"],
+        capture_output=True, text=True, cwd=REPO, timeout=240)
+    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+    assert "Eval acc" in r.stdout
