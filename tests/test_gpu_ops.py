@@ -374,28 +374,6 @@ def test_gather_rows_kernel(dev):
     assert torch.equal(ext.gather_rows(f3, ids, None, 0), f3)
 
 
-def test_bench_capture_mode_gpu(dev):
-    """--capture (whole-step hipGraph replay) produces a sane bench line."""
-    import json
-    import os
-    import subprocess
-    import sys
-
-    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-    r = subprocess.run(
-        [sys.executable, "bench.py", "--capture", "--steps", "4",
-         "--warmup", "2", "--nodes", "20000", "--edges", "150000"],
-        capture_output=True, text=True, cwd=repo, timeout=600,
-    )
-    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
-    d = json.loads([l for l in r.stdout.splitlines() if l.startswith("{")][-1])
-    assert d["value"] > 0
-    # edges per step must be plausible: ~batch * (25 + 10*frontier)-ish, and
-    # device-side accounting must not count the padded garbage region
-    edges_per_step = d["value"] * d["ms_per_step"] / 1000.0
-    assert 1000 < edges_per_step < 1000 * 36
-
-
 def test_gather_mm_mfma(dev):
     """fp32 MFMA gather-GEMM vs CPU reference. ASYMMETRIC weight (guide G9:
     symmetric operands miss transposed fragment layouts)."""
@@ -439,3 +417,25 @@ def test_sageconv_gatherview_gpu_matches_cpu(dev):
                           layer.fc_neigh.weight.grad, atol=1e-2, rtol=1e-2)
     assert torch.allclose(layer_gpu.fc_self.weight.grad.cpu(),
                           layer.fc_self.weight.grad, atol=1e-2, rtol=1e-2)
+
+
+def test_bench_capture_mode_gpu(dev):
+    """--capture (whole-step hipGraph replay) produces a sane bench line."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--capture", "--steps", "4",
+         "--warmup", "2", "--nodes", "20000", "--edges", "150000"],
+        capture_output=True, text=True, cwd=repo, timeout=600,
+    )
+    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+    d = json.loads([l for l in r.stdout.splitlines() if l.startswith("{")][-1])
+    assert d["value"] > 0
+    # edges per step must be plausible: ~batch * (25 + 10*frontier)-ish, and
+    # device-side accounting must not count the padded garbage region
+    edges_per_step = d["value"] * d["ms_per_step"] / 1000.0
+    assert 1000 < edges_per_step < 1000 * 36
