@@ -264,6 +264,8 @@ def main():
         edges = float(ee.cpu()[0])
 
     if rank == 0:
+        if device.type == "cuda":
+            print(f"# peak_mem_gb {torch.cuda.max_memory_allocated() / 1e9:.2f}")
         value = edges / elapsed
         print(json.dumps({
             "metric": "edges/sec (whole node) GraphSAGE ogbn-products",

@@ -100,7 +100,8 @@ class DistGraph:
         mask = (dst >= lo) & (dst < hi)
         psrc, pdst = src[mask], dst[mask] - lo
         indptr, indices, _ = _coo_to_compressed(pdst, psrc, hi - lo)
-        ndata = {k: v[lo:hi] for k, v in g.ndata.items()}
+        # clone the shard views so the full-graph tensors can be freed
+        ndata = {k: v[lo:hi].clone() for k, v in g.ndata.items()}
         return DistGraph(book, rank, indptr, indices, ndata)
 
     @staticmethod
