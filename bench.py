@@ -55,10 +55,10 @@ def parse_args():
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--profile", type=str, default="",
                    help="write a torch.profiler chrome trace of 3 steps here")
-    p.add_argument("--capture", action="store_true",
-                   help="hipGraph-capture the whole train step (1-GPU runs): "
-                        "worst-case shapes, zero host syncs, one replay per "
-                        "step")
+    p.add_argument("--capture", action="store_true", default=None,
+                   help="force hipGraph capture of the whole train step")
+    p.add_argument("--no-capture", dest="capture", action="store_false",
+                   help="disable the capture attempt (eager stepping)")
     return p.parse_args()
 
 
@@ -118,7 +118,10 @@ def main():
     if ws > 1:
         for p in model.parameters():
             dist.broadcast(p.data, src=0)
-    use_capture = args.capture and ws == 1 and device.type == "cuda"
+    # capture by default on 1-GPU CUDA runs; build failures fall back to
+    # eager stepping (a sanity replay checks the captured step first)
+    want_capture = (args.capture if args.capture is not None else True)
+    use_capture = want_capture and ws == 1 and device.type == "cuda"
     opt = torch.optim.Adam(model.parameters(), lr=args.lr,
                            capturable=use_capture)
 
@@ -161,6 +164,7 @@ def main():
     # Worst-case shapes keep every tensor static; actual sizes live on
     # device (see ops.sampling.sample_block_capture).
     if use_capture:
+      try:
         from dgl_operator_amd.ops.sampling import sample_block_capture
 
         static_seeds = torch.zeros(args.batch, dtype=torch.int64,
@@ -213,12 +217,27 @@ def main():
         graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(graph):
             capture_body()
+        # sanity: two replays must accumulate edges and keep weights finite
         edge_accum.zero_()
+        for s in range(2):
+            fill_seeds(90_000 + s)
+            graph.replay()
+        torch.cuda.synchronize()
+        assert float(edge_accum[0]) > 0, "captured step counted no edges"
+        for p_ in model.parameters():
+            assert bool(torch.isfinite(p_).all()), "non-finite weights"
+        if rank == 0:
+            print("# capture: enabled (hipGraph whole-step replay)")
 
         def one_step(step: int) -> int:
             fill_seeds(step)
             graph.replay()
             return 0  # edges tracked on device in edge_accum
+      except Exception as e:  # noqa: BLE001
+        use_capture = False
+        print(f"# capture: disabled ({type(e).__name__}: {e}); eager stepping")
+        # rebuild a non-capturable optimizer state
+        opt = torch.optim.Adam(model.parameters(), lr=args.lr)
 
     # warmup
     for s in range(args.warmup):

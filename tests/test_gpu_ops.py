@@ -433,6 +433,7 @@ def test_bench_capture_mode_gpu(dev):
         capture_output=True, text=True, cwd=repo, timeout=600,
     )
     assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+    assert "# capture: enabled" in r.stdout, r.stdout
     d = json.loads([l for l in r.stdout.splitlines() if l.startswith("{")][-1])
     assert d["value"] > 0
     # edges per step must be plausible: ~batch * (25 + 10*frontier)-ish, and
