@@ -121,7 +121,10 @@ def main():
     # capture by default on 1-GPU CUDA runs; build failures fall back to
     # eager stepping (a sanity replay checks the captured step first)
     want_capture = (args.capture if args.capture is not None else True)
-    use_capture = want_capture and ws == 1 and device.type == "cuda"
+    # capture's device-side valid-edge accounting is exact for the 2-layer
+    # flagship config; deeper samplers fall back to eager stepping
+    use_capture = (want_capture and ws == 1 and device.type == "cuda"
+                   and len(fanouts) == 2)
     opt = torch.optim.Adam(model.parameters(), lr=args.lr,
                            capturable=use_capture)
 
