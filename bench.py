@@ -307,10 +307,13 @@ def main():
                 "global_batch": args.batch * ws,
                 "seq_len": None,
                 "parallelism": (
+                    "dp1 (single GPU)" if ws == 1 else
                     f"dp{ws} (graph-partition data parallel, "
-                    + ("ghost-zone halo replication)" if ws > 1 and not args.no_halo
+                    + ("ghost-zone halo replication)" if not args.no_halo
                        else "alltoallv halo pulls)")
                 ),
+                "step_mode": ("hipGraph-captured" if use_capture
+                              else "eager"),
             },
         }))
 
