@@ -48,7 +48,9 @@ def parse_args():
     p.add_argument("--lr", type=float, default=3e-3)
     p.add_argument("--dropout", type=float, default=0.5)
     p.add_argument("--partition", type=str, default="range",
-                   help="range|ldg partition strategy for multi-rank runs")
+                   choices=["range"],
+                   help="in-bench sharding is contiguous ranges; LDG layouts "
+                        "come from the partitioner pod flow (graph/partition.py)")
     p.add_argument("--no-halo", action="store_true",
                    help="disable ghost-zone replication (fall back to "
                         "per-step alltoallv sampling + feature pulls)")
