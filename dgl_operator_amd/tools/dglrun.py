@@ -81,8 +81,21 @@ def run_phase1(args, dataset):
 
 def _deliver(args, dataset):
     with phase("Phase 2/5 deliver"):
-        with open(args.leadfile) as f:
-            leads = parse_hostfile(f.read())
+        # the leadfile is populated by the operator once the launcher pod has
+        # an IP; wait for it like the watcher-loop waits for pods
+        leads = []
+        deadline = time.time() + 120
+        while time.time() < deadline:
+            try:
+                with open(args.leadfile) as f:
+                    leads = parse_hostfile(f.read())
+            except FileNotFoundError:
+                leads = []
+            if leads:
+                break
+            time.sleep(0.5)
+        if not leads:
+            raise SystemExit("[dglrun] no launcher entry in leadfile")
         fabric = get_fabric()
         for lead in leads:
             # copy into the launcher's still-running watcher-loop-partitioner
