@@ -18,6 +18,13 @@ at::Tensor gather_rows(at::Tensor feat, at::Tensor gids,
                        c10::optional<at::Tensor> map, int64_t offset);
 at::Tensor gather_mm(at::Tensor feat, at::Tensor rows, at::Tensor weight,
                      c10::optional<at::Tensor> bias);
+at::Tensor gat_score_fwd(at::Tensor src, at::Tensor dst, at::Tensor el,
+                         at::Tensor er, double slope);
+std::tuple<at::Tensor, at::Tensor> gat_score_bwd(at::Tensor src,
+                                                 at::Tensor dst,
+                                                 at::Tensor el, at::Tensor er,
+                                                 at::Tensor gout,
+                                                 double slope);
 std::tuple<at::Tensor, at::Tensor> sample_neighbors(at::Tensor indptr,
                                                     at::Tensor indices,
                                                     at::Tensor seeds,
@@ -68,6 +75,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("map") = py::none(), py::arg("offset") = 0);
   m.def("gather_mm", &doa::gather_mm, py::arg("feat"), py::arg("rows"),
         py::arg("weight"), py::arg("bias") = py::none());
+  m.def("gat_score_fwd", &doa::gat_score_fwd);
+  m.def("gat_score_bwd", &doa::gat_score_bwd);
   m.def("sample_neighbors", &doa::sample_neighbors);
   m.def("compact_ids", &doa::compact_ids);
   m.def("sample_block", &doa::sample_block, py::arg("indptr"),

@@ -626,3 +626,94 @@ at::Tensor gather_rows(at::Tensor feat, at::Tensor gids,
 }
 
 }  // namespace doa
+
+namespace doa {
+
+// ---------------------------------------------------------------------------
+// Fused GAT attention score (K4-family): s[p,h] = LeakyReLU(el[src[p],h] +
+// er[dst[p],h]) over CSC positions — one kernel instead of the
+// gather + gather + add + leaky chain; backward scatters into el/er with
+// fp32 atomics (the sign is recomputed from el/er, nothing extra saved).
+// ---------------------------------------------------------------------------
+__global__ void gat_score_fwd_kernel(const int64_t* __restrict__ src,
+                                     const int64_t* __restrict__ dst,
+                                     const float* __restrict__ el,
+                                     const float* __restrict__ er,
+                                     float* __restrict__ out, int64_t E,
+                                     int H, float slope) {
+  const int64_t total = E * H;
+  for (int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       tid < total; tid += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t p = tid / H;
+    const int h = (int)(tid % H);
+    const float v = el[src[p] * H + h] + er[dst[p] * H + h];
+    out[tid] = v > 0.f ? v : slope * v;
+  }
+}
+
+__global__ void gat_score_bwd_kernel(const int64_t* __restrict__ src,
+                                     const int64_t* __restrict__ dst,
+                                     const float* __restrict__ el,
+                                     const float* __restrict__ er,
+                                     const float* __restrict__ gout,
+                                     float* __restrict__ gel,
+                                     float* __restrict__ ger, int64_t E,
+                                     int H, float slope) {
+  const int64_t total = E * H;
+  for (int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       tid < total; tid += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t p = tid / H;
+    const int h = (int)(tid % H);
+    const float v = el[src[p] * H + h] + er[dst[p] * H + h];
+    const float g = gout[tid] * (v > 0.f ? 1.f : slope);
+    atomicAdd(&gel[src[p] * H + h], g);
+    atomicAdd(&ger[dst[p] * H + h], g);
+  }
+}
+
+at::Tensor gat_score_fwd(at::Tensor src, at::Tensor dst, at::Tensor el,
+                         at::Tensor er, double slope) {
+  TORCH_CHECK(el.is_cuda() && el.scalar_type() == at::kFloat,
+              "gat_score: fp32 GPU tensors expected");
+  auto elc = el.contiguous();
+  auto erc = er.contiguous();
+  const int64_t E = src.numel();
+  const int H = elc.dim() > 1 ? elc.size(1) : 1;
+  auto out = (H == 1) ? at::empty({E}, elc.options())
+                      : at::empty({E, H}, elc.options());
+  const int block = 256;
+  auto stream = cur_stream();
+  hipLaunchKernelGGL(gat_score_fwd_kernel, dim3(grid_for(E * H, block)),
+                     dim3(block), 0, stream, src.data_ptr<int64_t>(),
+                     dst.data_ptr<int64_t>(), elc.data_ptr<float>(),
+                     erc.data_ptr<float>(), out.data_ptr<float>(), E, H,
+                     (float)slope);
+  DOA_CHECK_HIP(hipGetLastError());
+  return out;
+}
+
+std::tuple<at::Tensor, at::Tensor> gat_score_bwd(at::Tensor src,
+                                                 at::Tensor dst,
+                                                 at::Tensor el, at::Tensor er,
+                                                 at::Tensor gout,
+                                                 double slope) {
+  auto elc = el.contiguous();
+  auto erc = er.contiguous();
+  auto g = gout.contiguous();
+  const int64_t E = src.numel();
+  const int H = elc.dim() > 1 ? elc.size(1) : 1;
+  auto gel = at::zeros_like(elc);
+  auto ger = at::zeros_like(erc);
+  const int block = 256;
+  auto stream = cur_stream();
+  hipLaunchKernelGGL(gat_score_bwd_kernel, dim3(grid_for(E * H, block)),
+                     dim3(block), 0, stream, src.data_ptr<int64_t>(),
+                     dst.data_ptr<int64_t>(), elc.data_ptr<float>(),
+                     erc.data_ptr<float>(), g.data_ptr<float>(),
+                     gel.data_ptr<float>(), ger.data_ptr<float>(), E, H,
+                     (float)slope);
+  DOA_CHECK_HIP(hipGetLastError());
+  return std::make_tuple(gel, ger);
+}
+
+}  // namespace doa

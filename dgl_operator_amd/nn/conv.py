@@ -21,6 +21,7 @@ import torch.nn.functional as F
 
 from ..graph.graph import Block, Graph
 from ..ops import gspmm, edge_softmax_csc
+from ..ops.sddmm import gat_score
 from ..ops.gather_mm import GatherView
 
 
@@ -168,9 +169,8 @@ class GATConv(nn.Module):
         z = self.fc(x).view(-1, H, D)  # [N, H, D]
         el = (z * self.attn_l).sum(-1)  # [N, H]
         er = (z * self.attn_r).sum(-1)
-        _, indices, _ = g.csc()
-        dst = g.csc_dst()
-        score = self.leaky(el[indices] + er[dst])  # [E, H], csc order
+        # fused u_add_v + LeakyReLU attention logits (csc order)
+        score = gat_score(g, el, er, self.leaky.negative_slope)
         alpha = edge_softmax_csc(g, score)
         out = gspmm(g, "u_mul_e", "sum", z, _csc_weight(g, alpha))  # [Nd, H, D]
         out = out.reshape(-1, H * D)

@@ -1,6 +1,6 @@
 from . import backend
 from .spmm import gspmm, spmm_raw
-from .sddmm import sddmm_dot, edge_softmax, edge_softmax_csc
+from .sddmm import sddmm_dot, edge_softmax, edge_softmax_csc, gat_score
 from .segment import segment_reduce, mean_nodes
 from .sampling import (
     sample_neighbors,
@@ -20,6 +20,7 @@ __all__ = [
     "sddmm_dot",
     "edge_softmax",
     "edge_softmax_csc",
+    "gat_score",
     "segment_reduce",
     "mean_nodes",
     "sample_neighbors",

@@ -120,3 +120,44 @@ def edge_softmax_csc(gstruct, scores_csc: torch.Tensor) -> torch.Tensor:
     """Softmax over in-edge segments with scores ALREADY in CSC order (fast path)."""
     indptr, _, _ = gstruct.csc()
     return _EdgeSoftmax.apply(indptr, scores_csc)
+
+
+class _GATScore(torch.autograd.Function):
+    """Fused LeakyReLU(el[src] + er[dst]) over CSC positions (u_add_v +
+    activation, the GAT attention logits). GPU: one kernel each way
+    (csrc gat_score_*); CPU: the equivalent torch expression."""
+
+    @staticmethod
+    def forward(ctx, src, dst, el, er, slope):
+        ctx.save_for_backward(src, dst, el, er)
+        ctx.slope = slope
+        if el.is_cuda and el.dtype == torch.float32 and backend.has_extension():
+            ext = backend.ext_for(el)
+            return ext.gat_score_fwd(src, dst, el, er, slope)
+        v = el[src] + er[dst]
+        return torch.where(v > 0, v, slope * v)
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        src, dst, el, er = ctx.saved_tensors
+        grad_out = grad_out.contiguous()
+        if el.is_cuda and el.dtype == torch.float32 and backend.has_extension():
+            ext = backend.ext_for(el)
+            gel, ger = ext.gat_score_bwd(src, dst, el, er, grad_out, ctx.slope)
+            return None, None, gel, ger, None
+        v = el[src] + er[dst]
+        g = grad_out * torch.where(v > 0, torch.ones_like(v),
+                                   torch.full_like(v, ctx.slope))
+        gel = torch.zeros_like(el)
+        gel.index_add_(0, src, g)
+        ger = torch.zeros_like(er)
+        ger.index_add_(0, dst, g)
+        return None, None, gel, ger, None
+
+
+def gat_score(gstruct, el: torch.Tensor, er: torch.Tensor,
+              negative_slope: float = 0.2) -> torch.Tensor:
+    """Per-edge attention logits in CSC order: LeakyReLU(el_u + er_v)."""
+    _, indices, _ = gstruct.csc()
+    dst = gstruct.csc_dst()
+    return _GATScore.apply(indices, dst, el, er, negative_slope)

@@ -440,3 +440,26 @@ def test_bench_capture_mode_gpu(dev):
     # device-side accounting must not count the padded garbage region
     edges_per_step = d["value"] * d["ms_per_step"] / 1000.0
     assert 1000 < edges_per_step < 1000 * 36
+
+
+def test_gat_score_fused_gpu(dev, big_graph):
+    from dgl_operator_amd.ops import gat_score
+
+    g = big_graph
+    H = 4
+    el = torch.randn(g.num_nodes, H)
+    er = torch.randn(g.num_nodes, H)
+    el_c = el.clone().requires_grad_(True)
+    er_c = er.clone().requires_grad_(True)
+    s_cpu = gat_score(g, el_c, er_c, 0.2)
+    gout = torch.randn_like(s_cpu)
+    s_cpu.backward(gout)
+
+    gg = g.to(dev)
+    el_g = el.to(dev).requires_grad_(True)
+    er_g = er.to(dev).requires_grad_(True)
+    s_gpu = gat_score(gg, el_g, er_g, 0.2)
+    s_gpu.backward(gout.to(dev))
+    assert torch.allclose(s_gpu.cpu(), s_cpu, atol=1e-5, rtol=1e-5)
+    assert torch.allclose(el_g.grad.cpu(), el_c.grad, atol=1e-3, rtol=1e-3)
+    assert torch.allclose(er_g.grad.cpu(), er_c.grad, atol=1e-3, rtol=1e-3)

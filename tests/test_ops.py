@@ -298,3 +298,25 @@ def test_sageconv_gatherview_matches_dense():
     o2.sum().backward()
     assert layer.fc_neigh.weight.grad is not None
     assert layer.fc_self.weight.grad is not None
+
+
+def test_gat_score_fused_cpu():
+    from dgl_operator_amd.ops import gat_score
+
+    g = dense_graph()
+    el = torch.randn(g.num_nodes, 4, requires_grad=True)
+    er = torch.randn(g.num_nodes, 4, requires_grad=True)
+    s = gat_score(g, el, er, 0.2)
+    indptr, indices, _ = g.csc()
+    dst = g.csc_dst()
+    v = el.detach()[indices] + er.detach()[dst]
+    ref = torch.where(v > 0, v, 0.2 * v)
+    assert torch.allclose(s, ref, atol=1e-6)
+    gout = torch.randn_like(s)
+    s.backward(gout)
+    el2 = el.detach().clone().requires_grad_(True)
+    er2 = er.detach().clone().requires_grad_(True)
+    v2 = el2[indices] + er2[dst]
+    torch.where(v2 > 0, v2, 0.2 * v2).backward(gout)
+    assert torch.allclose(el.grad, el2.grad, atol=1e-5)
+    assert torch.allclose(er.grad, er2.grad, atol=1e-5)
