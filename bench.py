@@ -78,6 +78,79 @@ def flat_allreduce_grads(model):
         off += n
 
 
+def _build_captured_step(args, dg, model, opt, device, fanouts, next_seeds,
+                         rank):
+    """Capture the whole train step as a hipGraph; returns the replay fn and
+    exposes the device-side edge accumulator as an attribute."""
+    from dgl_operator_amd.ops.gather_mm import GatherView
+    from dgl_operator_amd.ops.sampling import sample_block_capture
+
+    static_seeds = torch.zeros(args.batch, dtype=torch.int64, device=device)
+    seed_dev = torch.zeros(1, dtype=torch.int64, device=device)
+    edge_accum = torch.zeros(1, dtype=torch.float64, device=device)
+    feat_t, label_t = dg.ndata["feat"], dg.ndata["label"]
+
+    def capture_body():
+        cur = static_seeds
+        blocks, counters = [], []
+        for layer, fanout in enumerate(reversed(fanouts)):
+            blk, ctr = sample_block_capture(
+                dg.csc_indptr, dg.csc_indices, dg.workspace, cur, fanout,
+                7777 + layer, seed_dev,
+            )
+            blocks.insert(0, blk)
+            counters.insert(0, ctr)
+            cur = blk.srcdata_nids
+        x = GatherView(feat_t, cur)
+        y = label_t[static_seeds]
+        loss = F.cross_entropy(model(blocks, x), y)
+        opt.zero_grad(set_to_none=False)
+        loss.backward()
+        opt.step()
+        # valid-edge metric: outer block fully valid; the inner block's
+        # valid dst prefix = actual src count of the outer block
+        v = torch.full((1,), args.batch, dtype=torch.int64, device=device)
+        for blk, ctr in zip(reversed(blocks), reversed(counters)):
+            edge_accum.add_(blk.csc_indptr[v[0]].to(torch.float64))
+            v = v + ctr
+        return loss
+
+    def fill_seeds(step):
+        static_seeds.copy_(next_seeds())
+        seed_dev.fill_(step + 1)
+
+    # warm up (allocates grads + Adam state) on a side stream, then capture
+    side = torch.cuda.Stream()
+    side.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(side):
+        for s in range(3):
+            fill_seeds(s)
+            capture_body()
+    torch.cuda.current_stream().wait_stream(side)
+    graph = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(graph):
+        capture_body()
+    # sanity: two replays must accumulate edges and keep weights finite
+    edge_accum.zero_()
+    for s in range(2):
+        fill_seeds(90_000 + s)
+        graph.replay()
+    torch.cuda.synchronize()
+    assert float(edge_accum[0]) > 0, "captured step counted no edges"
+    for p_ in model.parameters():
+        assert bool(torch.isfinite(p_).all()), "non-finite weights"
+    if rank == 0:
+        print("# capture: enabled (hipGraph whole-step replay)")
+
+    def one_step(step: int) -> int:
+        fill_seeds(step)
+        graph.replay()
+        return 0  # edges tracked on device in edge_accum
+
+    one_step.edge_accum = edge_accum
+    return one_step
+
+
 def main():
     args = parse_args()
     from dgl_operator_amd.distributed import DistGraph, PartitionBook, comm
@@ -164,85 +237,20 @@ def main():
         opt.step()
         return sum(b.num_edges for b in blocks)
 
-    # -- hipGraph-captured step (1-GPU): the whole train step — sampling,
-    # compaction, gather, fwd, bwd, Adam — replays as ONE graph launch.
-    # Worst-case shapes keep every tensor static; actual sizes live on
-    # device (see ops.sampling.sample_block_capture).
+    # -- hipGraph-captured step (1-GPU): sampling, compaction, gather, fwd,
+    # bwd and Adam replay as ONE graph launch. Worst-case shapes keep every
+    # tensor static; actual sizes live on device
+    # (ops.sampling.sample_block_capture). Failures fall back to eager.
     if use_capture:
-      try:
-        from dgl_operator_amd.ops.sampling import sample_block_capture
-
-        static_seeds = torch.zeros(args.batch, dtype=torch.int64,
-                                   device=device)
-        seed_dev = torch.zeros(1, dtype=torch.int64, device=device)
-        edge_accum = torch.zeros(1, dtype=torch.float64, device=device)
-        feat_t, label_t = dg.ndata["feat"], dg.ndata["label"]
-
-        def capture_body():
-            cur = static_seeds
-            blocks, counters = [], []
-            for layer, fanout in enumerate(reversed(fanouts)):
-                blk, ctr = sample_block_capture(
-                    dg.csc_indptr, dg.csc_indices, dg.workspace, cur,
-                    fanout, 7777 + layer, seed_dev,
-                )
-                blocks.insert(0, blk)
-                counters.insert(0, ctr)
-                cur = blk.srcdata_nids
-            from dgl_operator_amd.ops.gather_mm import GatherView
-
-            x = GatherView(feat_t, cur)
-            y = label_t[static_seeds]
-            logits = model(blocks, x)
-            loss = F.cross_entropy(logits, y)
-            opt.zero_grad(set_to_none=False)
-            loss.backward()
-            opt.step()
-            # valid-edge metric: outer block fully valid; each inner block's
-            # valid dst prefix = actual src count of the next-outer block
-            v = torch.full((1,), args.batch, dtype=torch.int64, device=device)
-            for blk, ctr in zip(reversed(blocks), reversed(counters)):
-                edge_accum.add_(blk.csc_indptr[v[0]].to(torch.float64))
-                v = v + ctr
-            return loss
-
-        def fill_seeds(step):
-            static_seeds.copy_(next_seeds())
-            seed_dev.fill_(step + 1)
-
-        # warm up (allocates grads, Adam state) on a side stream, then capture
-        side = torch.cuda.Stream()
-        side.wait_stream(torch.cuda.current_stream())
-        with torch.cuda.stream(side):
-            for s in range(3):
-                fill_seeds(s)
-                capture_body()
-        torch.cuda.current_stream().wait_stream(side)
-        edge_accum.zero_()
-        graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(graph):
-            capture_body()
-        # sanity: two replays must accumulate edges and keep weights finite
-        edge_accum.zero_()
-        for s in range(2):
-            fill_seeds(90_000 + s)
-            graph.replay()
-        torch.cuda.synchronize()
-        assert float(edge_accum[0]) > 0, "captured step counted no edges"
-        for p_ in model.parameters():
-            assert bool(torch.isfinite(p_).all()), "non-finite weights"
-        if rank == 0:
-            print("# capture: enabled (hipGraph whole-step replay)")
-
-        def one_step(step: int) -> int:
-            fill_seeds(step)
-            graph.replay()
-            return 0  # edges tracked on device in edge_accum
-      except Exception as e:  # noqa: BLE001
-        use_capture = False
-        print(f"# capture: disabled ({type(e).__name__}: {e}); eager stepping")
-        # rebuild a non-capturable optimizer state
-        opt = torch.optim.Adam(model.parameters(), lr=args.lr)
+        try:
+            one_step = _build_captured_step(
+                args, dg, model, opt, device, fanouts, next_seeds, rank
+            )
+        except Exception as e:  # noqa: BLE001
+            use_capture = False
+            print(f"# capture: disabled ({type(e).__name__}: {e}); "
+                  "eager stepping")
+            opt = torch.optim.Adam(model.parameters(), lr=args.lr)
 
     # warmup
     for s in range(args.warmup):
@@ -263,7 +271,7 @@ def main():
     if device.type == "cuda":
         torch.cuda.synchronize()
     if use_capture:
-        edge_accum.zero_()  # count only the timed steps
+        one_step.edge_accum.zero_()  # count only the timed steps
     comm.barrier()
     t0 = time.perf_counter()
     edges = 0
@@ -274,7 +282,7 @@ def main():
     comm.barrier()
     elapsed = time.perf_counter() - t0
     if use_capture:
-        edges = float(edge_accum.cpu()[0])
+        edges = float(one_step.edge_accum.cpu()[0])
 
     t = torch.tensor([elapsed], dtype=torch.float64)
     e = torch.tensor([edges], dtype=torch.float64)
