@@ -75,3 +75,55 @@ def test_dgljob_end_to_end_local_runtime(tmp_path):
         wd = tmp_path / f"e2e-sage-worker-{i}" / "workspace"
         assert (wd / "workload" / f"part{i}" / "graph.pt").exists()
         assert (wd / "hostfile_revised").exists()
+
+
+KE_MANIFEST = """
+apiVersion: qihoo.net/v1alpha1
+kind: DGLJob
+metadata:
+  name: e2e-ke
+  namespace: default
+spec:
+  partitionMode: Skip
+  cleanPodPolicy: Running
+  dglReplicaSpecs:
+    Launcher:
+      replicas: 1
+      template:
+        spec:
+          containers:
+          - name: launcher
+            image: local
+            command: ["python", "{repo}/examples/dgl_ke/train_ke.py"]
+            args:
+            - --model-name=ComplEx
+            - --hidden-dim=16
+            - --batch-size=64
+            - --neg-sample-size=8
+            - --max-step=20
+            - --log-interval=10
+            - --num-entities=3000
+            - --num-relations=10
+            - --num-triples=10000
+            - --save-path=ckpts
+    Worker:
+      replicas: 1
+      template:
+        spec:
+          containers:
+          - name: worker
+            image: local
+"""
+
+
+@pytest.mark.timeout(300)
+def test_dgljob_ke_skip_mode_local_runtime(tmp_path):
+    """Skip-mode DGLJob: launcher runs the KE workload directly (the
+    reference's Launcher_Workload path) and checkpoints shards."""
+    manifest = KE_MANIFEST.format(repo=REPO)
+    job = run_job(manifest, str(tmp_path), timeout=240,
+                  extra_env={"PYTHONPATH": REPO})
+    assert job.status.phase == JobPhase.COMPLETED, job.status
+    ck = tmp_path / "e2e-ke-launcher" / "ckpts"
+    assert (ck / "entity_shard0.pt").exists()
+    assert (ck / "relation_shard0.pt").exists()
