@@ -61,6 +61,10 @@ def parse_args():
                    help="force hipGraph capture of the whole train step")
     p.add_argument("--no-capture", dest="capture", action="store_false",
                    help="disable the capture attempt (eager stepping)")
+    p.add_argument("--capture-dist", action="store_true",
+                   help="EXPERIMENTAL: capture multi-rank steps too (halo "
+                        "mode; the graph records the RCCL all-reduce). "
+                        "Requires uniform capture success across ranks.")
     return p.parse_args()
 
 
@@ -88,24 +92,36 @@ def _build_captured_step(args, dg, model, opt, device, fanouts, next_seeds,
     static_seeds = torch.zeros(args.batch, dtype=torch.int64, device=device)
     seed_dev = torch.zeros(1, dtype=torch.int64, device=device)
     edge_accum = torch.zeros(1, dtype=torch.float64, device=device)
-    feat_t, label_t = dg.ndata["feat"], dg.ndata["label"]
+    if dg.halo is not None:
+        # halo mode: sample over the extended structure (global-id keyed),
+        # gather features through the halo map
+        feat_t, label_t = dg.halo.feats["feat"], dg.halo.feats["label"]
+        row_map, feat_map = dg.halo.row_map, dg.halo.feat_map
+        samp_indptr, samp_indices = dg.halo.indptr, dg.halo.indices
+    else:
+        feat_t, label_t = dg.ndata["feat"], dg.ndata["label"]
+        row_map = feat_map = None
+        samp_indptr, samp_indices = dg.csc_indptr, dg.csc_indices
 
     def capture_body():
         cur = static_seeds
         blocks, counters = [], []
         for layer, fanout in enumerate(reversed(fanouts)):
             blk, ctr = sample_block_capture(
-                dg.csc_indptr, dg.csc_indices, dg.workspace, cur, fanout,
+                samp_indptr, samp_indices, dg.workspace, cur, fanout,
                 7777 + layer, seed_dev,
+                rows=row_map[cur] if row_map is not None else None,
             )
             blocks.insert(0, blk)
             counters.insert(0, ctr)
             cur = blk.srcdata_nids
-        x = GatherView(feat_t, cur)
-        y = label_t[static_seeds]
+        x = GatherView(feat_t, feat_map[cur] if feat_map is not None else cur)
+        y = label_t[feat_map[static_seeds] if feat_map is not None
+                    else static_seeds]
         loss = F.cross_entropy(model(blocks, x), y)
         opt.zero_grad(set_to_none=False)
         loss.backward()
+        flat_allreduce_grads(model)  # recorded into the graph when ws > 1
         opt.step()
         # valid-edge metric: outer block fully valid; the inner block's
         # valid dst prefix = actual src count of the outer block
@@ -198,8 +214,9 @@ def main():
     want_capture = (args.capture if args.capture is not None else True)
     # capture's device-side valid-edge accounting is exact for the 2-layer
     # flagship config; deeper samplers fall back to eager stepping
-    use_capture = (want_capture and ws == 1 and device.type == "cuda"
-                   and len(fanouts) == 2)
+    multi_ok = (ws > 1 and args.capture_dist and not args.no_halo)
+    use_capture = (want_capture and (ws == 1 or multi_ok)
+                   and device.type == "cuda" and len(fanouts) == 2)
     opt = torch.optim.Adam(model.parameters(), lr=args.lr,
                            capturable=use_capture)
 

@@ -157,6 +157,7 @@ def sample_block_capture(
     fanout: int,
     seed: int,
     seed_dev: torch.Tensor,
+    rows: "torch.Tensor | None" = None,
 ) -> "tuple[Block, torch.Tensor]":
     """hipGraph-capturable block build: NO host synchronization. Shapes are
     worst-case (num_src = n + n*fanout, packed buffer n*fanout); the actual
@@ -169,7 +170,8 @@ def sample_block_capture(
     claimed source nodes (needed for the valid-edge metric)."""
     ext = backend.ext_for(seeds)
     padded, counts, srcdata, counter = ext.sample_block(
-        indptr, indices, workspace.table, seeds, fanout, False, seed, seed_dev
+        indptr, indices, workspace.table, seeds, fanout, False, seed, seed_dev,
+        rows,
     )
     n = seeds.numel()
     blk_indptr = torch.zeros(n + 1, dtype=torch.int64, device=seeds.device)

@@ -463,3 +463,20 @@ def test_gat_score_fused_gpu(dev, big_graph):
     assert torch.allclose(s_gpu.cpu(), s_cpu, atol=1e-5, rtol=1e-5)
     assert torch.allclose(el_g.grad.cpu(), el_c.grad, atol=1e-3, rtol=1e-3)
     assert torch.allclose(er_g.grad.cpu(), er_c.grad, atol=1e-3, rtol=1e-3)
+
+
+def test_bf16_sddmm_and_segment(dev, big_graph):
+    """Low-precision paths of the remaining aggregation kernels."""
+    g = big_graph
+    src, dst = g.edges()
+    fu = torch.randn(g.num_nodes, 64)
+    ref = (fu[src] * fu[dst]).sum(-1)
+    out = sddmm_dot_raw(src.to(dev), dst.to(dev),
+                        fu.to(dev).bfloat16(), fu.to(dev).bfloat16())
+    assert out.dtype == torch.bfloat16
+    assert torch.allclose(out.float().cpu(), ref, atol=2.0, rtol=0.05)
+    offsets = torch.tensor([0, 10, 500, 1000])
+    x = torch.randn(1000, 32)
+    ref = _segment_ref(offsets, x, True)
+    out = segment_reduce(offsets.to(dev), x.to(dev).bfloat16(), "mean")
+    assert torch.allclose(out.float().cpu(), ref, atol=0.1, rtol=0.05)
