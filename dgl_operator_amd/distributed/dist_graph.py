@@ -385,8 +385,9 @@ def _build_halo_cache(dg: "DistGraph", num_layers: int,
     feat_map[feat_ids] = torch.arange(feat_ids.numel(), device=dev) + n_owned
     comm.barrier()
     for key in feat_keys:
-        halo_rows = dg.pull(key, feat_ids) if feat_ids.numel() else \
-            dg.ndata[key][:0]
+        # NOTE: pull is a symmetric collective — every rank must join even
+        # with an empty request list (others may be requesting from us)
+        halo_rows = dg.pull(key, feat_ids)
         cache.feats[key] = torch.cat([dg.ndata[key], halo_rows])
     cache.indptr = ext_indptr
     cache.indices = ext_indices
