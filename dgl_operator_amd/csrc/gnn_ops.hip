@@ -442,6 +442,7 @@ at::Tensor edge_softmax_fwd(at::Tensor indptr, at::Tensor scores) {
   auto s = scores.contiguous();
   const int64_t num_rows = indptr.numel() - 1;
   const int64_t E = s.size(0);
+  if (E == 0) return s.clone();
   const int H = (s.dim() > 1) ? (int)(s.numel() / E) : 1;
   auto out = at::empty_like(s);
   const int block = 256;
@@ -468,6 +469,7 @@ at::Tensor edge_softmax_bwd(at::Tensor indptr, at::Tensor out,
   auto g = grad_out.contiguous();
   const int64_t num_rows = indptr.numel() - 1;
   const int64_t E = a.size(0);
+  if (E == 0) return a.clone();
   const int H = (a.dim() > 1) ? (int)(a.numel() / E) : 1;
   auto gin = at::empty_like(a);
   const int block = 256;
