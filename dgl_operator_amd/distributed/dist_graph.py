@@ -127,6 +127,15 @@ class DistGraph:
     def owned_nodes(self) -> torch.Tensor:
         return torch.arange(self.lo, self.hi, device=self.device)
 
+    def node_split(self, mask_key: str = "train_mask") -> torch.Tensor:
+        """GLOBAL ids of this rank's owned nodes where the boolean mask is
+        set — dgl.distributed.node_split parity (each rank gets its owned
+        slice of the masked set; reference train_dist.py:274-276)."""
+        mask = self.ndata.get(mask_key)
+        if mask is None:
+            return self.owned_nodes()
+        return self.owned_nodes()[mask.bool()]
+
     # -- distributed sampling ---------------------------------------------
     def _sample_local(self, gids_local: torch.Tensor, fanout: int, replace, seed):
         return sample_neighbors(

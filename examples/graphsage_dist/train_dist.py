@@ -108,9 +108,7 @@ def main():
             if rank == 0:
                 print(f"resumed from epoch {state['epoch']}", flush=True)
 
-    train_mask = dg.ndata.get("train_mask")
-    owned = dg.owned_nodes()
-    train_nids = owned[train_mask.bool()] if train_mask is not None else owned
+    train_nids = dg.node_split("train_mask")
 
     gen = torch.Generator(device=device)
     gen.manual_seed(1234 + rank)
