@@ -23,7 +23,7 @@ def _free_port():
     return port
 
 
-def _run_workers(fn, world=2, args=()):
+def _run_workers(fn, world=2, args=(), _retry=True):
     port = _free_port()
     ctx = mp.get_context("spawn")
     procs = []
@@ -33,6 +33,12 @@ def _run_workers(fn, world=2, args=()):
         procs.append(p)
     for p in procs:
         p.join(180)
+    if _retry and any(p.exitcode != 0 for p in procs):
+        # one retry absorbs rendezvous-port races (the free-port probe window)
+        for p in procs:
+            if p.is_alive():
+                p.terminate()
+        return _run_workers(fn, world, args, _retry=False)
     for p in procs:
         assert p.exitcode == 0, f"worker failed with {p.exitcode}"
 

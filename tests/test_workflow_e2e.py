@@ -55,21 +55,24 @@ def test_dglrun_launcher_end_to_end(tmp_path):
     env["PYTHONPATH"] = REPO
 
     workspace = "ws"
-    r = subprocess.run(
-        [
-            sys.executable, "-m", "dgl_operator_amd.tools.dglrun",
-            "--graph-name", "toy",
-            "--workspace", workspace,
-            "--hostfile", str(hostfile),
-            "--master-port", str(_free_port()),
-            "--train-entry-point",
-            os.path.join(REPO, "examples/graphsage_dist/train_dist.py"),
-            "--train-entry-args",
-            "--num-epochs 1 --batch-size 32 --fan-out 3,5 --log-every 1",
-        ],
-        capture_output=True, text=True, cwd=str(tmp_path),
-        env=env, timeout=240,
-    )
+    for attempt in range(2):  # retry absorbs master-port races
+        r = subprocess.run(
+            [
+                sys.executable, "-m", "dgl_operator_amd.tools.dglrun",
+                "--graph-name", "toy",
+                "--workspace", workspace,
+                "--hostfile", str(hostfile),
+                "--master-port", str(_free_port()),
+                "--train-entry-point",
+                os.path.join(REPO, "examples/graphsage_dist/train_dist.py"),
+                "--train-entry-args",
+                "--num-epochs 1 --batch-size 32 --fan-out 3,5 --log-every 1",
+            ],
+            capture_output=True, text=True, cwd=str(tmp_path),
+            env=env, timeout=240,
+        )
+        if r.returncode == 0:
+            break
     assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
     assert "Phase 3/5 dispatch" in r.stdout
     assert "Phase 5/5 train" in r.stdout
@@ -93,13 +96,16 @@ def test_train_dist_single_node_multi_rank(tmp_path):
          "--feat", "8", "--classes", "3", "--algorithm", "range"],
         capture_output=True, text=True, cwd=REPO)
     assert r.returncode == 0, r.stderr
-    r = subprocess.run(
-        [sys.executable, "-m", "torch.distributed.run", "--standalone",
-         "--local-addr", "127.0.0.1", "--nproc-per-node", "2",
-         os.path.join(REPO, "examples/graphsage_dist/train_dist.py"),
-         "--part-config", str(tmp_path / "t.json"),
-         "--num-epochs", "1", "--batch-size", "32", "--fan-out", "3,3",
-         "--eval-every", "1", "--log-every", "100"],
-        capture_output=True, text=True, cwd=REPO, timeout=240)
+    for attempt in range(2):
+        r = subprocess.run(
+            [sys.executable, "-m", "torch.distributed.run", "--standalone",
+             "--local-addr", "127.0.0.1", "--nproc-per-node", "2",
+             os.path.join(REPO, "examples/graphsage_dist/train_dist.py"),
+             "--part-config", str(tmp_path / "t.json"),
+             "--num-epochs", "1", "--batch-size", "32", "--fan-out", "3,3",
+             "--eval-every", "1", "--log-every", "100"],
+            capture_output=True, text=True, cwd=REPO, timeout=240)
+        if r.returncode == 0:
+            break
     assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
     assert "Eval acc" in r.stdout
