@@ -419,29 +419,6 @@ def test_sageconv_gatherview_gpu_matches_cpu(dev):
                           layer.fc_self.weight.grad, atol=1e-2, rtol=1e-2)
 
 
-def test_bench_capture_mode_gpu(dev):
-    """--capture (whole-step hipGraph replay) produces a sane bench line."""
-    import json
-    import os
-    import subprocess
-    import sys
-
-    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-    r = subprocess.run(
-        [sys.executable, "bench.py", "--capture", "--steps", "4",
-         "--warmup", "2", "--nodes", "20000", "--edges", "150000"],
-        capture_output=True, text=True, cwd=repo, timeout=600,
-    )
-    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
-    assert "# capture: enabled" in r.stdout, r.stdout
-    d = json.loads([l for l in r.stdout.splitlines() if l.startswith("{")][-1])
-    assert d["value"] > 0
-    # edges per step must be plausible: ~batch * (25 + 10*frontier)-ish, and
-    # device-side accounting must not count the padded garbage region
-    edges_per_step = d["value"] * d["ms_per_step"] / 1000.0
-    assert 1000 < edges_per_step < 1000 * 36
-
-
 def test_gat_score_fused_gpu(dev, big_graph):
     from dgl_operator_amd.ops import gat_score
 
@@ -480,3 +457,26 @@ def test_bf16_sddmm_and_segment(dev, big_graph):
     ref = _segment_ref(offsets, x, True)
     out = segment_reduce(offsets.to(dev), x.to(dev).bfloat16(), "mean")
     assert torch.allclose(out.float().cpu(), ref, atol=0.1, rtol=0.05)
+
+
+def test_bench_capture_mode_gpu(dev):
+    """--capture (whole-step hipGraph replay) produces a sane bench line."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--capture", "--steps", "4",
+         "--warmup", "2", "--nodes", "20000", "--edges", "150000"],
+        capture_output=True, text=True, cwd=repo, timeout=600,
+    )
+    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+    assert "# capture: enabled" in r.stdout, r.stdout
+    d = json.loads([l for l in r.stdout.splitlines() if l.startswith("{")][-1])
+    assert d["value"] > 0
+    # edges per step must be plausible: ~batch * (25 + 10*frontier)-ish, and
+    # device-side accounting must not count the padded garbage region
+    edges_per_step = d["value"] * d["ms_per_step"] / 1000.0
+    assert 1000 < edges_per_step < 1000 * 36
