@@ -11,6 +11,7 @@
 
 #include <torch/extension.h>
 #include <ATen/Parallel.h>
+#include <limits>
 #include <c10/hip/HIPStream.h>
 
 #include "common.h"
@@ -31,7 +32,8 @@ template <typename scalar_t, int VEC, WeightMode WM>
 __global__ void spmm_kernel(
     const int64_t* __restrict__ indptr, const int64_t* __restrict__ indices,
     const scalar_t* __restrict__ feat, const scalar_t* __restrict__ ew,
-    scalar_t* __restrict__ out, int64_t num_rows, int F, int D, bool mean) {
+    scalar_t* __restrict__ out, int64_t num_rows, int F, int D, bool mean,
+    int64_t thresh) {
   const int chunks = F / VEC;
   const int64_t total = num_rows * chunks;
   for (int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; tid < total;
@@ -41,6 +43,7 @@ __global__ void spmm_kernel(
     const int f0 = c * VEC;
     const int h = (WM == W_HEAD) ? (f0 / D) : 0;
     const int64_t p0 = indptr[row], p1 = indptr[row + 1];
+    if (p1 - p0 > thresh) continue;  // hub rows: block-parallel kernel
     using acc_t = typename AccT<scalar_t>::type;
     acc_t acc[VEC];
 #pragma unroll
@@ -79,6 +82,57 @@ __global__ void spmm_kernel(
   }
 }
 
+// Hub rows (power-law full-graph aggregation / layer-wise inference):
+// one 256-thread block per long row; threads split (neighbor stripe x
+// feature chunk), partial sums reduced through an LDS tile. Parallelizes
+// the neighbor loop 256/chunks-way instead of serializing 10k+ iterations
+// per thread.
+constexpr int64_t kSpmmLongRow = 256;
+
+template <typename scalar_t, int VEC, WeightMode WM>
+__global__ void spmm_long_kernel(
+    const int64_t* __restrict__ indptr, const int64_t* __restrict__ indices,
+    const scalar_t* __restrict__ feat, const scalar_t* __restrict__ ew,
+    scalar_t* __restrict__ out, int64_t num_rows, int F, int D, bool mean,
+    int64_t thresh) {
+  using acc_t = typename AccT<scalar_t>::type;
+  extern __shared__ float sacc[];  // [F] fp32 partial tile
+  const int chunks = F / VEC;
+  const int tpc = blockDim.x / chunks;  // threads cooperating per chunk
+  for (int64_t row = blockIdx.x; row < num_rows; row += gridDim.x) {
+    const int64_t p0 = indptr[row], p1 = indptr[row + 1];
+    const int64_t deg = p1 - p0;
+    if (deg <= thresh) continue;
+    for (int i = threadIdx.x; i < F; i += blockDim.x) sacc[i] = 0.f;
+    __syncthreads();
+    const int c = threadIdx.x % chunks;
+    const int stripe = threadIdx.x / chunks;
+    const int f0 = c * VEC;
+    const int h = (WM == W_HEAD) ? (f0 / D) : 0;
+    if (stripe < tpc) {
+      float acc[VEC];
+#pragma unroll
+      for (int i = 0; i < VEC; ++i) acc[i] = 0.f;
+      for (int64_t p = p0 + stripe; p < p1; p += tpc) {
+        const int64_t u = indices[p];
+        float w = 1.f;
+        if (WM == W_SCALAR) w = (float)ew[p];
+        if (WM == W_HEAD) w = (float)ew[p * (F / D) + h];
+        const scalar_t* src = feat + u * F + f0;
+#pragma unroll
+        for (int i = 0; i < VEC; ++i) acc[i] += w * (float)src[i];
+      }
+#pragma unroll
+      for (int i = 0; i < VEC; ++i) atomicAdd(&sacc[f0 + i], acc[i]);
+    }
+    __syncthreads();
+    const float inv = mean ? 1.f / (float)deg : 1.f;
+    for (int i = threadIdx.x; i < F; i += blockDim.x)
+      out[row * F + i] = (scalar_t)(sacc[i] * inv);
+    __syncthreads();
+  }
+}
+
 template <typename scalar_t>
 static void spmm_launch(const at::Tensor& indptr, const at::Tensor& indices,
                         const at::Tensor& feat, const c10::optional<at::Tensor>& ew,
@@ -97,12 +151,27 @@ static void spmm_launch(const at::Tensor& indptr, const at::Tensor& indices,
                     (wm != W_HEAD || (D % 8 == 0));
   const int chunks = vec4 ? F / 4 : (vec8 ? F / 8 : F);
   const int grid = grid_for(num_rows * chunks, block);
+  // the hub-row kernel needs >=1 thread per feature chunk and an F-float
+  // LDS tile; outside that envelope the per-thread kernel takes every row
+  const bool long_ok =
+      (chunks <= block) && ((size_t)F * sizeof(float) <= 64 * 1024);
+  const int64_t thresh =
+      long_ok ? kSpmmLongRow : std::numeric_limits<int64_t>::max();
   auto stream = cur_stream();
 #define DOA_SPMM(V, W)                                                        \
   hipLaunchKernelGGL((spmm_kernel<scalar_t, V, W>), dim3(grid), dim3(block),  \
                      0, stream, indptr.data_ptr<int64_t>(),                   \
                      indices.data_ptr<int64_t>(), feat.data_ptr<scalar_t>(),  \
-                     ewp, out.data_ptr<scalar_t>(), num_rows, F, D, mean)
+                     ewp, out.data_ptr<scalar_t>(), num_rows, F, D, mean,     \
+                     thresh);                                                 \
+  if (long_ok)                                                                \
+    hipLaunchKernelGGL((spmm_long_kernel<scalar_t, V, W>),                    \
+                       dim3(grid_for(num_rows * 256, block)), dim3(block),    \
+                       F * sizeof(float), stream, indptr.data_ptr<int64_t>(), \
+                       indices.data_ptr<int64_t>(),                           \
+                       feat.data_ptr<scalar_t>(), ewp,                        \
+                       out.data_ptr<scalar_t>(), num_rows, F, D, mean,        \
+                       kSpmmLongRow)
   if (vec4) {
     if (wm == W_NONE) DOA_SPMM(4, W_NONE);
     else if (wm == W_SCALAR) DOA_SPMM(4, W_SCALAR);

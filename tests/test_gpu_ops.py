@@ -480,3 +480,34 @@ def test_bench_capture_mode_gpu(dev):
     # device-side accounting must not count the padded garbage region
     edges_per_step = d["value"] * d["ms_per_step"] / 1000.0
     assert 1000 < edges_per_step < 1000 * 36
+
+
+def test_spmm_hub_rows(dev):
+    """Hub rows (> 256 in-edges) go through the block-parallel LDS kernel;
+    outputs must match the CPU reference including the short/long boundary."""
+    torch.manual_seed(2)
+    # star graph: node 0 has 5000 in-edges; plus a band of short rows
+    hub_src = torch.randint(1, 4000, (5000,))
+    hub_dst = torch.zeros(5000, dtype=torch.int64)
+    band_src = torch.randint(0, 4000, (8000,))
+    band_dst = torch.randint(1, 4000, (8000,))
+    from dgl_operator_amd.graph import Graph
+
+    g = Graph(torch.cat([hub_src, band_src]), torch.cat([hub_dst, band_dst]),
+              4000)
+    indptr, indices, _ = g.csc()
+    for F in [16, 100, 1024, 7]:
+        x = torch.randn(4000, F)
+        for mean in (False, True):
+            ref = _spmm_ref(indptr, indices, x, None, mean)
+            out = spmm_raw(indptr.to(dev), indices.to(dev), x.to(dev), None,
+                           mean)
+            assert torch.allclose(out.cpu(), ref, atol=1e-3, rtol=1e-3), (
+                F, mean)
+    # weighted hub
+    x = torch.randn(4000, 32)
+    w = torch.rand(g.num_edges)
+    ref = _spmm_ref(indptr, indices, x, w, True)
+    out = spmm_raw(indptr.to(dev), indices.to(dev), x.to(dev), w.to(dev),
+                   True)
+    assert torch.allclose(out.cpu(), ref, atol=1e-3, rtol=1e-3)
