@@ -459,29 +459,6 @@ def test_bf16_sddmm_and_segment(dev, big_graph):
     assert torch.allclose(out.float().cpu(), ref, atol=0.1, rtol=0.05)
 
 
-def test_bench_capture_mode_gpu(dev):
-    """--capture (whole-step hipGraph replay) produces a sane bench line."""
-    import json
-    import os
-    import subprocess
-    import sys
-
-    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-    r = subprocess.run(
-        [sys.executable, "bench.py", "--capture", "--steps", "4",
-         "--warmup", "2", "--nodes", "20000", "--edges", "150000"],
-        capture_output=True, text=True, cwd=repo, timeout=600,
-    )
-    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
-    assert "# capture: enabled" in r.stdout, r.stdout
-    d = json.loads([l for l in r.stdout.splitlines() if l.startswith("{")][-1])
-    assert d["value"] > 0
-    # edges per step must be plausible: ~batch * (25 + 10*frontier)-ish, and
-    # device-side accounting must not count the padded garbage region
-    edges_per_step = d["value"] * d["ms_per_step"] / 1000.0
-    assert 1000 < edges_per_step < 1000 * 36
-
-
 def test_spmm_hub_rows(dev):
     """Hub rows (> 256 in-edges) go through the block-parallel LDS kernel;
     outputs must match the CPU reference including the short/long boundary."""
@@ -511,3 +488,26 @@ def test_spmm_hub_rows(dev):
     out = spmm_raw(indptr.to(dev), indices.to(dev), x.to(dev), w.to(dev),
                    True)
     assert torch.allclose(out.cpu(), ref, atol=1e-3, rtol=1e-3)
+
+
+def test_bench_capture_mode_gpu(dev):
+    """--capture (whole-step hipGraph replay) produces a sane bench line."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--capture", "--steps", "4",
+         "--warmup", "2", "--nodes", "20000", "--edges", "150000"],
+        capture_output=True, text=True, cwd=repo, timeout=600,
+    )
+    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+    assert "# capture: enabled" in r.stdout, r.stdout
+    d = json.loads([l for l in r.stdout.splitlines() if l.startswith("{")][-1])
+    assert d["value"] > 0
+    # edges per step must be plausible: ~batch * (25 + 10*frontier)-ish, and
+    # device-side accounting must not count the padded garbage region
+    edges_per_step = d["value"] * d["ms_per_step"] / 1000.0
+    assert 1000 < edges_per_step < 1000 * 36
