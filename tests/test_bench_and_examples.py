@@ -56,6 +56,16 @@ def test_example_workloads_run(script, args):
 
 
 @pytest.mark.timeout(300)
+def _free_port_2():
+    import socket
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
 def test_dglkerun_end_to_end(tmp_path):
     hostfile = tmp_path / "hostfile"
     hostfile.write_text(
@@ -68,19 +78,23 @@ def test_dglkerun_end_to_end(tmp_path):
     s.bind(("127.0.0.1", 0))
     port = s.getsockname()[1]
     s.close()
-    r = _run(
-        [sys.executable, "-m", "dgl_operator_amd.tools.dglkerun",
-         "--hostfile", str(hostfile),
-         "--workspace", "ws",
-         "--master-port", str(port),
-         "--train-entry-point", os.path.join(REPO, "examples/dgl_ke/train_ke.py"),
-         "--model-name", "TransE_l2", "--hidden-dim", "16",
-         "--batch-size", "64", "--neg-sample-size", "8", "--max-step", "20",
-         "--save-path", "ckpts"],
-        env={"DGL_LOCAL_FABRIC_ROOT": str(tmp_path / "pods"),
-             "PYTHONPATH": REPO},
-        cwd=str(tmp_path),
-    )
+    for attempt in range(2):  # retry absorbs master-port races
+        r = _run(
+            [sys.executable, "-m", "dgl_operator_amd.tools.dglkerun",
+             "--hostfile", str(hostfile),
+             "--workspace", "ws",
+             "--master-port", str(port if attempt == 0 else _free_port_2()),
+             "--train-entry-point",
+             os.path.join(REPO, "examples/dgl_ke/train_ke.py"),
+             "--model-name", "TransE_l2", "--hidden-dim", "16",
+             "--batch-size", "64", "--neg-sample-size", "8", "--max-step", "20",
+             "--save-path", "ckpts"],
+            env={"DGL_LOCAL_FABRIC_ROOT": str(tmp_path / "pods"),
+                 "PYTHONPATH": REPO},
+            cwd=str(tmp_path),
+        )
+        if r.returncode == 0:
+            break
     assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
     assert "Phase 5/5 dglke train" in r.stdout
     # sharded checkpoints written per pod per rank
