@@ -55,7 +55,6 @@ def test_example_workloads_run(script, args):
     assert r.returncode == 0, f"{script}: {r.stderr[-2000:]}"
 
 
-@pytest.mark.timeout(300)
 def _free_port_2():
     import socket
 
@@ -66,6 +65,7 @@ def _free_port_2():
     return port
 
 
+@pytest.mark.timeout(300)
 def test_dglkerun_end_to_end(tmp_path):
     hostfile = tmp_path / "hostfile"
     hostfile.write_text(
