@@ -43,7 +43,9 @@ void sparse_adagrad(at::Tensor emb, at::Tensor state, at::Tensor ids,
                     at::Tensor grad, double lr, double eps);
 at::Tensor ldg_partition(at::Tensor indptr, at::Tensor indices,
                          at::Tensor cindptr, at::Tensor cindices,
-                         int64_t num_parts);
+                         int64_t num_parts,
+                         c10::optional<at::Tensor> train_mask,
+                         bool balance_edges);
 at::Tensor pdist_neg_fwd(at::Tensor base, at::Tensor neg, int64_t p,
                          double gamma);
 std::tuple<at::Tensor, at::Tensor> pdist_neg_bwd(at::Tensor base,
@@ -85,7 +87,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("seed_dev") = py::none(), py::arg("rows") = py::none());
   m.def("pack_padded", &doa::pack_padded);
   m.def("sparse_adagrad", &doa::sparse_adagrad);
-  m.def("ldg_partition", &doa::ldg_partition);
+  m.def("ldg_partition", &doa::ldg_partition, py::arg("indptr"),
+        py::arg("indices"), py::arg("cindptr"), py::arg("cindices"),
+        py::arg("num_parts"), py::arg("train_mask") = py::none(),
+        py::arg("balance_edges") = false);
   m.def("pdist_neg_fwd", &doa::pdist_neg_fwd);
   m.def("pdist_neg_bwd", &doa::pdist_neg_bwd);
   m.def("cpdist_neg_fwd", &doa::cpdist_neg_fwd);

@@ -22,7 +22,9 @@ static inline uint64_t mix64(uint64_t x) {
 
 at::Tensor ldg_partition(at::Tensor indptr, at::Tensor indices,
                          at::Tensor cindptr, at::Tensor cindices,
-                         int64_t num_parts) {
+                         int64_t num_parts,
+                         c10::optional<at::Tensor> train_mask,
+                         bool balance_edges) {
   TORCH_CHECK(!indptr.is_cuda(), "ldg_partition runs on CPU tensors");
   const int64_t n = indptr.numel() - 1;
   const int64_t* ip = indptr.data_ptr<int64_t>();
@@ -41,8 +43,24 @@ at::Tensor ldg_partition(at::Tensor indptr, at::Tensor indices,
     std::swap(order[i], order[j]);
   }
 
+  // balance dimensions (the reference's METIS balance_train/balance_edges,
+  // load_and_partition_graph.py:124-127): node count always; optionally the
+  // train-node count and the in+out edge count
+  const bool* tm = nullptr;
+  at::Tensor tmc;
+  if (train_mask.has_value()) {
+    tmc = train_mask->to(at::kBool).contiguous();
+    tm = tmc.data_ptr<bool>();
+  }
   const double cap = (double)(n + num_parts - 1) / num_parts * 1.05 + 1.0;
+  double train_total = 0.0, edge_total = 0.0;
+  if (tm) for (int64_t i = 0; i < n; ++i) train_total += tm[i] ? 1.0 : 0.0;
+  if (balance_edges)
+    edge_total = (double)(indices.numel() + cindices.numel());
+  const double tcap = train_total / num_parts * 1.05 + 1.0;
+  const double ecap = edge_total / num_parts * 1.05 + 1.0;
   std::vector<int64_t> sizes(num_parts, 0);
+  std::vector<double> tsizes(num_parts, 0.0), esizes(num_parts, 0.0);
   std::vector<int64_t> counts(num_parts);
   for (int64_t t = 0; t < n; ++t) {
     const int64_t v = order[t];
@@ -55,10 +73,15 @@ at::Tensor ldg_partition(at::Tensor indptr, at::Tensor indices,
       const int64_t a = assign[cix[p]];
       if (a >= 0) counts[a]++;
     }
+    const double vdeg =
+        (double)((ip[v + 1] - ip[v]) + (cip[v + 1] - cip[v]));
     int best = 0;
     double best_score = -1.0;
     for (int64_t q = 0; q < num_parts; ++q) {
-      const double score = (counts[q] + 1e-9) * (1.0 - sizes[q] / cap);
+      double penalty = 1.0 - sizes[q] / cap;
+      if (tm) penalty *= (1.0 - tsizes[q] / tcap);
+      if (balance_edges) penalty *= (1.0 - esizes[q] / ecap);
+      const double score = (counts[q] + 1e-9) * std::max(penalty, 0.0);
       if (score > best_score) {
         best_score = score;
         best = (int)q;
@@ -66,6 +89,8 @@ at::Tensor ldg_partition(at::Tensor indptr, at::Tensor indices,
     }
     assign[v] = best;
     sizes[best]++;
+    if (tm && tm[v]) tsizes[best] += 1.0;
+    if (balance_edges) esizes[best] += vdeg;
   }
   return out;
 }

@@ -106,7 +106,8 @@ def _assign_ldg_python(g: Graph, num_parts: int) -> torch.Tensor:
     return torch.tensor(assign, dtype=torch.int64)
 
 
-def _assign_ldg(g: Graph, num_parts: int) -> torch.Tensor:
+def _assign_ldg(g: Graph, num_parts: int, balance_train: bool = False,
+                balance_edges: bool = False) -> torch.Tensor:
     try:
         from ..ops import backend
 
@@ -114,8 +115,12 @@ def _assign_ldg(g: Graph, num_parts: int) -> torch.Tensor:
         if ext is not None and hasattr(ext, "ldg_partition"):
             indptr, indices, _ = g.csr()
             cindptr, cindices, _ = g.csc()
+            mask = g.ndata.get("train_mask") if balance_train else None
             return ext.ldg_partition(
-                indptr.cpu(), indices.cpu(), cindptr.cpu(), cindices.cpu(), num_parts
+                indptr.cpu(), indices.cpu(), cindptr.cpu(), cindices.cpu(),
+                num_parts,
+                mask.cpu() if mask is not None else None,
+                balance_edges,
             )
     except Exception:
         pass
@@ -129,15 +134,18 @@ def partition_graph(
     out_path: str,
     algorithm: str = "ldg",
     balance_train: bool = False,
+    balance_edges: bool = False,
 ) -> PartitionSpec:
     """Partition ``g`` and write the on-disk layout. Returns the metadata spec.
 
     ``g.ndata`` entries are split by ownership and stored per part.
+    ``balance_train``/``balance_edges`` mirror the reference's METIS balance
+    objectives (load_and_partition_graph.py:124-127) in the LDG scorer.
     """
     if algorithm == "range":
         assign = _assign_range(g.num_nodes, num_parts)
     elif algorithm == "ldg":
-        assign = _assign_ldg(g, num_parts)
+        assign = _assign_ldg(g, num_parts, balance_train, balance_edges)
     elif algorithm == "random":
         assign = torch.randint(
             0, num_parts, (g.num_nodes,), generator=torch.Generator().manual_seed(0)

@@ -110,3 +110,22 @@ def test_dglrun_skip_mode(tmp_path):
     assert r.returncode == 0, r.stderr
     assert "train-ran" in r.stdout
     assert "workload (Skip mode)" in r.stdout
+
+
+def test_partition_balance_objectives(tmp_path):
+    """balance_train / balance_edges keep the per-part train-node and edge
+    loads balanced (the reference's METIS balance objectives)."""
+    g = rmat_graph(400, 6000, seed=7)
+    gen = torch.Generator().manual_seed(1)
+    g.ndata["train_mask"] = torch.rand(400, generator=gen) < 0.2
+    spec = partition_graph(g, "bal", 4, str(tmp_path), algorithm="ldg",
+                           balance_train=True, balance_edges=True)
+    # per-part train counts within 2x of each other
+    tloads, eloads = [], []
+    for p in range(4):
+        gp, feats, _ = load_partition(str(tmp_path / "bal.json"), p)
+        tloads.append(int(feats["train_mask"].sum()))
+        eloads.append(int(gp["src_global"].numel()))
+    assert min(tloads) > 0
+    assert max(tloads) <= 2.5 * max(1, min(tloads)), tloads
+    assert max(eloads) <= 2.5 * max(1, min(eloads)), eloads
