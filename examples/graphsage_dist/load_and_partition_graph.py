@@ -32,6 +32,12 @@ def main():
     p.add_argument("--classes", type=int, default=47)
     p.add_argument("--algorithm", default="ldg", choices=["ldg", "range", "random"])
     p.add_argument("--train-fraction", type=float, default=0.1)
+    # the reference partitions with balance_ntypes=train + balance_edges
+    # (load_and_partition_graph.py:124-127)
+    p.add_argument("--no-balance-train", dest="balance_train",
+                   action="store_false")
+    p.add_argument("--no-balance-edges", dest="balance_edges",
+                   action="store_false")
     args = p.parse_args()
 
     from dgl_operator_amd.graph import Graph, partition_graph, rmat_graph
@@ -51,7 +57,9 @@ def main():
         torch.rand(g.num_nodes, generator=gen) < args.train_fraction
     )
     spec = partition_graph(g, args.graph_name, args.num_partitions, args.output,
-                           algorithm=args.algorithm)
+                           algorithm=args.algorithm,
+                           balance_train=args.balance_train,
+                           balance_edges=args.balance_edges)
     print(f"partitioned {args.graph_name}: {spec.num_parts} parts, "
           f"boundaries {spec.boundaries}")
 
