@@ -95,10 +95,16 @@ class Manager:
                         JOB_PHASE.labels(job=job.name, phase=ph.value).set(
                             1.0 if job.status.phase == ph else 0.0
                         )
-            except Exception:  # noqa: BLE001
+            except Exception as e:  # noqa: BLE001
+                # a failing job must not stall the loop for other jobs —
+                # the reference requeues on error (Reconcile returns err)
                 if _PROM:
                     RECONCILE_ERRORS.labels(job=job.name).inc()
-                raise
+                import traceback
+
+                print(f"[manager] reconcile error for {key}: {e}")
+                traceback.print_exc()
+                continue
             if job.deletion_timestamp is not None:
                 del self.jobs[key]
 

@@ -307,3 +307,15 @@ def test_shipped_manifests_parse():
             job = job_from_manifest(f.read())
         assert job.name
         assert job.num_workers() >= 1
+
+
+def test_manager_survives_bad_job():
+    from dgl_operator_amd.operator_plane.manager import Manager
+
+    mgr = Manager()
+    good = mgr.submit(GRAPHSAGE_YAML)
+    bad = mgr.submit(GRAPHSAGE_YAML.replace("graphsage-dist", "bad-job"))
+    # corrupt the bad job so reconcile raises
+    bad.spec.replica_specs = None
+    mgr.reconcile_once()  # must not raise; good job still progresses
+    assert mgr.cluster.get_pod("default", "graphsage-dist-launcher") is not None
