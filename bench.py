@@ -57,6 +57,9 @@ def parse_args():
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--profile", type=str, default="",
                    help="write a torch.profiler chrome trace of 3 steps here")
+    p.add_argument("--phase-timing", action="store_true",
+                   help="print a sample/gather/fwd/bwd/opt breakdown "
+                        "(eager mode, CUDA events)")
     p.add_argument("--capture", action="store_true", default=None,
                    help="force hipGraph capture of the whole train step")
     p.add_argument("--no-capture", dest="capture", action="store_false",
@@ -239,19 +242,41 @@ def main():
         cursor[0] += args.batch
         return s
 
+    phase_ms = {k: 0.0 for k in ("sample", "gather", "fwd", "bwd", "opt")}
+    phase_events = []
+    if args.phase_timing and device.type == "cuda":
+        phase_events = [torch.cuda.Event(enable_timing=True)
+                        for _ in range(6)]
+
     def one_step(step: int) -> int:
+        ev = phase_events
+        if ev:
+            ev[0].record()
         seeds = next_seeds()
         input_nodes, output_nodes, blocks = dg.sample_blocks(
             seeds, fanouts, seed=step + 1
         )
+        if ev:
+            ev[1].record()
         x = dg.pull_view("feat", input_nodes)
         y = dg.pull("label", output_nodes)
+        if ev:
+            ev[2].record()
         logits = model(blocks, x)
         loss = F.cross_entropy(logits, y)
+        if ev:
+            ev[3].record()
         opt.zero_grad(set_to_none=True)
         loss.backward()
         flat_allreduce_grads(model)
+        if ev:
+            ev[4].record()
         opt.step()
+        if ev:
+            ev[5].record()
+            torch.cuda.synchronize()
+            for i, k in enumerate(phase_ms):
+                phase_ms[k] += ev[i].elapsed_time(ev[i + 1])
         return sum(b.num_edges for b in blocks)
 
     # -- hipGraph-captured step (1-GPU): sampling, compaction, gather, fwd,
@@ -315,6 +340,11 @@ def main():
     if rank == 0:
         if device.type == "cuda":
             print(f"# peak_mem_gb {torch.cuda.max_memory_allocated() / 1e9:.2f}")
+        if args.phase_timing and phase_events and not use_capture:
+            total_steps = args.warmup + args.steps + (3 if args.profile else 0)
+            breakdown = {k: round(v / total_steps, 3)
+                         for k, v in phase_ms.items()}
+            print(f"# phase_ms_per_step {breakdown}")
         value = edges / elapsed
         print(json.dumps({
             "metric": "edges/sec (whole node) GraphSAGE ogbn-products",
