@@ -26,7 +26,7 @@ rocprofv3 --kernel-trace --stats -d "$OUT/bench_trace" -- \
 # 2. PMC counters for the SpMM/sampler kernels (separate run, counters only)
 rocprofv3 --pmc SQ_WAVES,SQ_BUSY_CYCLES,TCC_HIT_sum,TCC_MISS_sum \
     -d "$OUT/bench_pmc" -- \
-    timeout 300 python bench.py --steps 5 --warmup 2 > "$OUT/bench_pmc.log" 2>&1
+    timeout 300 python bench.py --steps 5 --warmup 2 --no-capture > "$OUT/bench_pmc.log" 2>&1
 
 # 3. kernel-trace of the KGE path
 rocprofv3 --kernel-trace --stats -d "$OUT/ke_trace" -- \
