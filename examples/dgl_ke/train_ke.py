@@ -23,6 +23,77 @@ import time
 import torch
 
 
+def _capture_ke_step(model, sampler, args, device):
+    """Build two captured KE steps (neg-tail / neg-head) over static id
+    buffers; returns a replay(step) callable exposing the static loss."""
+    import torch as th
+
+    from dgl_operator_amd.ops import kge_loss
+
+    B, C = args.batch_size, args.batch_size // args.chunk_size
+    s_h = th.zeros(B, dtype=th.int64, device=device)
+    s_r = th.zeros(B, dtype=th.int64, device=device)
+    s_t = th.zeros(B, dtype=th.int64, device=device)
+    s_n = th.zeros(C, args.neg_sample_size, dtype=th.int64, device=device)
+    losses = {}
+
+    def body(neg_head):
+        h = model.entities.pull(s_h).requires_grad_(True)
+        t = model.entities.pull(s_t).requires_grad_(True)
+        n = model.entities.pull(s_n.reshape(-1)).requires_grad_(True)
+        r = model.relations.pull(s_r).requires_grad_(True)
+        pos = model.score.edge(h, r, t)
+        hc = (t if neg_head else h).view(C, args.chunk_size, -1)
+        rc = r.view(C, args.chunk_size, -1)
+        nc = n.view(C, args.neg_sample_size, -1)
+        neg = model.score.neg(hc, rc, nc, neg_head=neg_head)
+        loss = kge_loss(pos, neg, 1.0)
+        loss.backward()
+        with th.no_grad():
+            ent_ids = th.cat([s_h, s_t, s_n.reshape(-1)])
+            ent_grad = th.cat([h.grad, t.grad, n.grad])
+            model.entities.push_grad(ent_ids, ent_grad, args.lr)
+            model.relations.push_grad(s_r, r.grad, args.lr)
+        losses[neg_head] = losses.get(neg_head, th.zeros((), device=device))
+        losses[neg_head].copy_(loss.detach())
+        return loss
+
+    def fill():
+        hh, rr, tt, negs, neg_head = sampler.next_batch()
+        s_h.copy_(hh)
+        s_r.copy_(rr)
+        s_t.copy_(tt)
+        s_n.copy_(negs)
+        return neg_head
+
+    graphs = {}
+    side = th.cuda.Stream()
+    side.wait_stream(th.cuda.current_stream())
+    with th.cuda.stream(side):
+        for _ in range(4):  # warm both corruption sides
+            body(fill())
+    th.cuda.current_stream().wait_stream(side)
+    for nh in (False, True):
+        g = th.cuda.CUDAGraph()
+        with th.cuda.graph(g):
+            body(nh)
+        graphs[nh] = g
+    # sanity replay both sides
+    for _ in range(2):
+        nh = fill()
+        graphs[nh].replay()
+    th.cuda.synchronize()
+    for nh in (False, True):
+        assert bool(th.isfinite(losses[nh])), "non-finite captured loss"
+
+    def replay(step):
+        nh = fill()
+        graphs[nh].replay()
+        return losses[nh]
+
+    return replay
+
+
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--model-name", default="ComplEx")
@@ -39,6 +110,10 @@ def main():
     p.add_argument("--num-triples", type=int, default=5_000_000)
     p.add_argument("--save-path", default="")
     p.add_argument("--no-save-emb", action="store_true")
+    p.add_argument("--no-capture", dest="capture", action="store_false",
+                   default=True,
+                   help="disable hipGraph capture of the train step "
+                        "(1-GPU runs capture by default; shapes are static)")
     p.add_argument("--json", action="store_true",
                    help="print a bench-style JSON line (triples/s) at the end")
     p.add_argument("--eval", action="store_true",
@@ -74,12 +149,32 @@ def main():
         neg_sample_size=args.neg_sample_size, chunk_size=args.chunk_size,
         seed=rank + 1, device=device,
     )
+    # hipGraph capture (1-GPU): the KE step is fully static-shaped — ids go
+    # into static buffers, one graph replay per step, loss read only at log
+    # points. Head/tail corruption alternates per step, so TWO graphs are
+    # captured (one per corruption side) and replayed alternately.
+    captured = None
+    if args.capture and ws == 1 and device.type == "cuda":
+        try:
+            captured = _capture_ke_step(model, sampler, args, device)
+            print("# capture: enabled (hipGraph KE step, 2 graphs)",
+                  flush=True)
+        except Exception as e:  # noqa: BLE001
+            print(f"# capture: disabled ({type(e).__name__}: {e})",
+                  flush=True)
+
     t0 = time.time()
     for step in range(1, args.max_step + 1):
-        hh, rr, tt, negs, neg_head = sampler.next_batch()
-        loss = model.train_step(hh, rr, tt, negs, args.chunk_size, args.lr,
-                                neg_head=neg_head)
+        if captured is not None:
+            loss_t = captured(step)
+            loss = None
+        else:
+            hh, rr, tt, negs, neg_head = sampler.next_batch()
+            loss = model.train_step(hh, rr, tt, negs, args.chunk_size,
+                                    args.lr, neg_head=neg_head)
         if step % args.log_interval == 0 and rank == 0:
+            if loss is None:
+                loss = float(loss_t.detach().cpu())
             elapsed = time.time() - t0
             tps = step * args.batch_size * ws / elapsed
             print(f"step {step} loss {loss:.4f} {tps:,.0f} triples/s",
