@@ -71,23 +71,26 @@ class DistKGEModel:
         num_chunk, n_neg = neg_entities.shape
         assert B == num_chunk * chunk_size
         dim = self.hidden_dim
-        h = self.entities.pull(heads).requires_grad_(True)
-        t = self.entities.pull(tails).requires_grad_(True)
-        n = self.entities.pull(neg_entities.reshape(-1)).requires_grad_(True)
+        # ONE entity pull/push round per step: h, t and the negatives share a
+        # single alltoallv (the reference pulls them separately through the
+        # kvstore — dis_kvstore.py:818-902)
+        ent_ids = torch.cat([heads, tails, neg_entities.reshape(-1)])
+        rows = self.entities.pull(ent_ids).requires_grad_(True)
+        h = rows[:B]
+        t = rows[B : 2 * B]
+        n = rows[2 * B :]
         r = self.relations.pull(rels).requires_grad_(True)
 
         pos = self.score.edge(h, r, t)
-        hc = (t if neg_head else h).view(num_chunk, chunk_size, dim)
+        hc = (t if neg_head else h).reshape(num_chunk, chunk_size, dim)
         rc = r.view(num_chunk, chunk_size, -1)
-        nc = n.view(num_chunk, n_neg, dim)
+        nc = n.reshape(num_chunk, n_neg, dim)
         neg = self.score.neg(hc, rc, nc, neg_head=neg_head)
         loss = kge_loss(pos, neg, adversarial_temperature)
         loss.backward()
 
         with torch.no_grad():
-            ent_ids = torch.cat([heads, tails, neg_entities.reshape(-1)])
-            ent_grad = torch.cat([h.grad, t.grad, n.grad])
-            self.entities.push_grad(ent_ids, ent_grad, lr)
+            self.entities.push_grad(ent_ids, rows.grad, lr)
             self.relations.push_grad(rels, r.grad, lr)
         return float(loss.detach())
 

@@ -38,21 +38,20 @@ def _capture_ke_step(model, sampler, args, device):
     losses = {}
 
     def body(neg_head):
-        h = model.entities.pull(s_h).requires_grad_(True)
-        t = model.entities.pull(s_t).requires_grad_(True)
-        n = model.entities.pull(s_n.reshape(-1)).requires_grad_(True)
+        ent_ids = th.cat([s_h, s_t, s_n.reshape(-1)])
+        rows = model.entities.pull(ent_ids).requires_grad_(True)
+        h, t = rows[:B], rows[B : 2 * B]
+        n = rows[2 * B :]
         r = model.relations.pull(s_r).requires_grad_(True)
         pos = model.score.edge(h, r, t)
-        hc = (t if neg_head else h).view(C, args.chunk_size, -1)
+        hc = (t if neg_head else h).reshape(C, args.chunk_size, -1)
         rc = r.view(C, args.chunk_size, -1)
-        nc = n.view(C, args.neg_sample_size, -1)
+        nc = n.reshape(C, args.neg_sample_size, -1)
         neg = model.score.neg(hc, rc, nc, neg_head=neg_head)
         loss = kge_loss(pos, neg, 1.0)
         loss.backward()
         with th.no_grad():
-            ent_ids = th.cat([s_h, s_t, s_n.reshape(-1)])
-            ent_grad = th.cat([h.grad, t.grad, n.grad])
-            model.entities.push_grad(ent_ids, ent_grad, args.lr)
+            model.entities.push_grad(ent_ids, rows.grad, args.lr)
             model.relations.push_grad(s_r, r.grad, args.lr)
         losses[neg_head] = losses.get(neg_head, th.zeros((), device=device))
         losses[neg_head].copy_(loss.detach())
