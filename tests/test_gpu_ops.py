@@ -361,7 +361,6 @@ def test_gather_rows_kernel(dev):
     out = ext.gather_rows(feat, gids, None, 50)
     assert torch.equal(out, feat[gids - 50])
     # map form
-    mp = torch.randperm(2000, device=dev)[:1050]
     mp_full = torch.full((2000,), -1, dtype=torch.int64, device=dev)
     mp_full[:1050] = torch.randint(0, 1000, (1050,), device=dev)
     out2 = ext.gather_rows(feat, gids.clamp(max=1049), mp_full, 0)
