@@ -57,6 +57,11 @@ class ShardedEmbedding:
     def pull(self, gids: torch.Tensor) -> torch.Tensor:
         rank, ws = comm.world()
         if ws == 1:
+            if gids.is_cuda and self.local.is_floating_point():
+                from ..ops import backend
+
+                ext = backend.ext_for(gids)
+                return ext.gather_rows(self.local, gids, None, self.lo)
             return self.local[gids - self.lo]
         sorted_ids, perm, send_counts = self.book.partition_by_owner(gids)
         recv_counts = comm.exchange_counts(send_counts)
