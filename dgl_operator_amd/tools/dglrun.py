@@ -39,15 +39,30 @@ def _run(cmd: str, env=None):
 
 
 def build_parser():
+    # flag surface parity with the reference CLI (python/dglrun/exec/dglrun
+    # arg parser; its misspelled --worksapce is accepted as an alias)
     p = argparse.ArgumentParser(prog="dglrun")
-    p.add_argument("--graph-name", default="graph")
+    p.add_argument("-g", "--graph-name", default="graph")
     p.add_argument("--partition-entry-point", default="")
     p.add_argument("--partition-entry-args", default="")
     p.add_argument("--num-partitions", type=int, default=1)
+    p.add_argument("--balance-train", action="store_true")
+    p.add_argument("--balance-edges", action="store_true")
     p.add_argument("--train-entry-point", default="")
     p.add_argument("--train-entry-args", default="")
-    p.add_argument("--workspace", default=os.environ.get("WORKSPACE",
-                                                         "/dgl_workspace"))
+    p.add_argument("--num-epochs", type=int, default=None,
+                   help="appended to the train args when set")
+    p.add_argument("--batch-size", type=int, default=None,
+                   help="appended to the train args when set")
+    p.add_argument("--partition-config-path", default="",
+                   help="override the part-config path handed to training")
+    p.add_argument("--dispatch-entry-point", default="",
+                   help="custom dispatch script (default: built-in dispatch)")
+    p.add_argument("--num-servers", type=int, default=0,
+                   help="accepted for reference-CLI parity; this framework "
+                        "has no server processes")
+    p.add_argument("--workspace", "--worksapce", default=os.environ.get(
+        "WORKSPACE", "/dgl_workspace"))
     p.add_argument("--hostfile", default="/etc/dgl/hostfile")
     p.add_argument("--leadfile", default="/etc/dgl/leadfile")
     p.add_argument("--master-port", type=int, default=29400)
@@ -71,11 +86,16 @@ def run_partitioner(args):
 
 def run_phase1(args, dataset):
     with phase("Phase 1/5 partition"):
+        extra = args.partition_entry_args
+        if not args.balance_train:
+            extra += " --no-balance-train"
+        if not args.balance_edges:
+            extra += " --no-balance-edges"
         _run(
             f"python {args.partition_entry_point} "
             f"--graph-name {args.graph_name} "
             f"--num-partitions {args.num_partitions} "
-            f"--output {dataset} {args.partition_entry_args}"
+            f"--output {dataset} {extra}"
         )
 
 
@@ -109,8 +129,13 @@ def run_launcher(args):
         hosts = parse_hostfile(f.read())
     dataset = os.path.join(args.workspace, "dataset")
     with phase("Phase 3/5 dispatch"):
-        dispatch_partitions(dataset, args.graph_name, hosts,
-                            workspace=args.workspace)
+        if args.dispatch_entry_point:
+            _run(f"python {args.dispatch_entry_point} "
+                 f"--dataset-dir {dataset} --graph-name {args.graph_name} "
+                 f"--hostfile {args.hostfile} --workspace {args.workspace}")
+        else:
+            dispatch_partitions(dataset, args.graph_name, hosts,
+                                workspace=args.workspace)
     with phase("Phase 4/5 revise hostfile"):
         revised = revise_for_dgl(hosts)
         launch_mod.exec_batch(
@@ -119,12 +144,18 @@ def run_launcher(args):
             f"> {args.workspace}/hostfile_revised'",
         )
     with phase("Phase 5/5 train"):
+        part_cfg = (args.partition_config_path
+                    or f"{args.workspace}/workload/{args.graph_name}.json")
+        targs = args.train_entry_args
+        if args.num_epochs is not None:
+            targs += f" --num-epochs {args.num_epochs}"
+        if args.batch_size is not None:
+            targs += f" --batch-size {args.batch_size}"
         launch_mod.train(
             hosts,
             args.train_entry_point,
             f"--graph-name {args.graph_name} "
-            f"--part-config {args.workspace}/workload/{args.graph_name}.json "
-            f"{args.train_entry_args}",
+            f"--part-config {part_cfg} {targs}",
             master_port=args.master_port,
         )
 
