@@ -28,8 +28,27 @@ def phase(name: str):
 
 
 def build_parser():
+    # flag surface parity with the reference CLI (python/dglrun/exec/dglkerun;
+    # --worksapce is its misspelling, accepted as an alias). Flags tied to the
+    # reference's server/client process architecture (--num-servers,
+    # --num-client-procs, --mix-cpus-and-single-gpu) are accepted and ignored:
+    # this framework runs one rank per GPU with a sharded kvstore.
     p = argparse.ArgumentParser(prog="dglkerun")
-    p.add_argument("--model-name", default="ComplEx")
+    p.add_argument("--model-name", "--model", default="ComplEx")
+    p.add_argument("--dataset", default="synthetic")
+    p.add_argument("--custom-dataset", default="")
+    p.add_argument("--dataset-files", default="")
+    p.add_argument("--dataset-format", default="")
+    p.add_argument("--num-partitions", type=int, default=0)
+    p.add_argument("--num-servers", type=int, default=0)
+    p.add_argument("--num-client-procs", type=int, default=0)
+    p.add_argument("--mix-cpus-and-single-gpu", action="store_true")
+    p.add_argument("--partition-config-path", default="")
+    p.add_argument("--pvc-partitioned-dir", default="",
+                   help="alias of --partitioned-dataset-dir (PVC reuse)")
+    p.add_argument("--dispatch-entry-point", default="")
+    p.add_argument("--launch-entry-point", default="")
+    p.add_argument("--revise-hostfile-entry-point", default="")
     p.add_argument("--hidden-dim", type=int, default=400)
     p.add_argument("--gamma", type=float, default=143.0)
     p.add_argument("--lr", type=float, default=0.1)
@@ -40,8 +59,8 @@ def build_parser():
     p.add_argument("--no-save-emb", action="store_true")
     p.add_argument("--ignore-partition", action="store_true")
     p.add_argument("--partitioned-dataset-dir", default="")
-    p.add_argument("--workspace", default=os.environ.get("WORKSPACE",
-                                                         "/dgl_workspace"))
+    p.add_argument("--workspace", "--worksapce", default=os.environ.get(
+        "WORKSPACE", "/dgl_workspace"))
     p.add_argument("--hostfile", default="/etc/dgl/hostfile")
     p.add_argument("--master-port", type=int, default=29401)
     p.add_argument("--train-entry-point",
