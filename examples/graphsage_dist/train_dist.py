@@ -29,18 +29,35 @@ import torch.nn.functional as F
 
 
 def parse_args():
+    # accepts both this repo's dashed flags and the reference train script's
+    # underscore spellings (train_dist.py:284-318); flags tied to the
+    # reference's process architecture (--num_workers sampler procs,
+    # --num_servers, --ip_config) are accepted and ignored.
     p = argparse.ArgumentParser(description="GraphSAGE distributed trainer")
-    p.add_argument("--graph-name", type=str, default="graph")
-    p.add_argument("--part-config", type=str, required=True)
-    p.add_argument("--num-epochs", type=int, default=1)
-    p.add_argument("--num-hidden", type=int, default=16)
-    p.add_argument("--num-layers", type=int, default=2)
-    p.add_argument("--fan-out", type=str, default="10,25")
-    p.add_argument("--batch-size", type=int, default=1000)
+    p.add_argument("--graph-name", "--graph_name", type=str, default="graph")
+    p.add_argument("--part-config", "--part_config", type=str, required=True)
+    p.add_argument("--id", type=int, default=None,
+                   help="partition id override (default: global rank)")
+    p.add_argument("--ip-config", "--ip_config", type=str, default="")
+    p.add_argument("--num-clients", "--num_clients", type=int, default=0)
+    p.add_argument("--num-servers", "--num_servers", type=int, default=0)
+    p.add_argument("--n-classes", "--n_classes", type=int, default=0)
+    p.add_argument("--num-gpus", "--num_gpus", type=int, default=None,
+                   help="-1 forces CPU training (reference semantics)")
+    p.add_argument("--num-epochs", "--num_epochs", type=int, default=1)
+    p.add_argument("--num-hidden", "--num_hidden", type=int, default=16)
+    p.add_argument("--num-layers", "--num_layers", type=int, default=2)
+    p.add_argument("--fan-out", "--fan_out", type=str, default="10,25")
+    p.add_argument("--batch-size", "--batch_size", type=int, default=1000)
+    p.add_argument("--batch-size-eval", "--batch_size_eval", type=int,
+                   default=100000)
     p.add_argument("--lr", type=float, default=0.003)
     p.add_argument("--dropout", type=float, default=0.5)
-    p.add_argument("--log-every", type=int, default=20)
-    p.add_argument("--eval-every", type=int, default=0)
+    p.add_argument("--log-every", "--log_every", type=int, default=20)
+    p.add_argument("--eval-every", "--eval_every", type=int, default=0)
+    p.add_argument("--num-workers", "--num_workers", type=int, default=0)
+    p.add_argument("--local-rank", "--local_rank", type=int, default=None)
+    p.add_argument("--standalone", action="store_true")
     p.add_argument("--checkpoint-path", type=str, default="",
                    help="save model+optimizer per epoch; resume if present")
     p.add_argument("--no-halo", action="store_true",
@@ -54,7 +71,8 @@ def main():
     from dgl_operator_amd.distributed import DistGraph, comm
 
     rank, ws = comm.init_from_env()
-    if torch.cuda.is_available():
+    force_cpu = args.num_gpus is not None and args.num_gpus == -1
+    if torch.cuda.is_available() and not force_cpu:
         device = torch.device(f"cuda:{int(os.environ.get('LOCAL_RANK', 0))}")
         torch.cuda.set_device(device)
     else:
@@ -63,7 +81,7 @@ def main():
     # one partition per RANK: the partition book maps global rank -> owned
     # range, so the job must be partitioned into world_size parts
     # (slotsPerWorker=1 deployments: rank == worker pod index)
-    part_id = rank
+    part_id = rank if args.id is None else args.id
     dg = DistGraph.from_partition(args.part_config, part_id, device=device)
     if ws > 1:
         assert dg.book.num_parts == ws, (
@@ -78,7 +96,7 @@ def main():
     feat = dg.ndata["feat"]
     in_feats = feat.shape[1]
     labels = dg.ndata["label"]
-    n_classes = int(labels.max().item()) + 1
+    n_classes = args.n_classes or int(labels.max().item()) + 1
     nc = torch.tensor([n_classes], device=device)
     if ws > 1:
         dist.all_reduce(nc, op=dist.ReduceOp.MAX)
@@ -158,7 +176,8 @@ def main():
         if args.eval_every and (epoch + 1) % args.eval_every == 0:
             from dgl_operator_amd.models.graphsage import inference_dist
 
-            logits_shard = inference_dist(model, dg, batch_size=4096)
+            logits_shard = inference_dist(
+                model, dg, batch_size=args.batch_size_eval)
             pred = logits_shard.argmax(1)
             labels_local = dg.ndata["label"]
             correct = (pred == labels_local).sum()
