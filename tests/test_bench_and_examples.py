@@ -45,6 +45,8 @@ def test_bench_json_contract():
      ["--epochs", "3", "--nodes", "300", "--edges", "2000"]),
     ("examples/graph_classification/train.py",
      ["--epochs", "2", "--num-graphs", "20"]),
+    ("examples/message_passing/train.py",
+     ["--epochs", "5", "--nodes", "300", "--edges", "2000", "--feat", "16"]),
     ("examples/dgl_ke/train_ke.py",
      ["--max-step", "10", "--log-interval", "5", "--num-entities", "2000",
       "--num-relations", "10", "--num-triples", "5000", "--hidden-dim", "16",
