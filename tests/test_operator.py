@@ -319,3 +319,46 @@ def test_manager_survives_bad_job():
     bad.spec.replica_specs = None
     mgr.reconcile_once()  # must not raise; good job still progresses
     assert mgr.cluster.get_pod("default", "graphsage-dist-launcher") is not None
+
+
+def test_kubectl_cluster_and_watcher_cli(tmp_path, monkeypatch):
+    """KubectlCluster parses pod JSON from a stubbed kubectl; the watcher-loop
+    CLI (the init-container binary equivalent) exits 0 once pods are ready."""
+    import json as _json
+    import subprocess
+    import sys
+
+    fake = tmp_path / "kubectl"
+    pod_json = {
+        "metadata": {"labels": {"dgl-job-name": "j"}},
+        "status": {
+            "phase": "Running",
+            "podIP": "10.0.0.5",
+            "containerStatuses": [{"ready": True}],
+        },
+    }
+    fake.write_text(
+        "#!/bin/sh\n"
+        f"echo '{_json.dumps(pod_json)}'\n"
+    )
+    fake.chmod(0o755)
+    monkeypatch.setenv("PATH", f"{tmp_path}:{__import__('os').environ['PATH']}")
+
+    from dgl_operator_amd.operator_plane.cluster import KubectlCluster
+
+    c = KubectlCluster()
+    p = c.get_pod("default", "j-worker-0")
+    assert p.phase.value == "Running" and p.ip == "10.0.0.5"
+    assert p.is_real_running()
+
+    watchfile = tmp_path / "hostfile"
+    watchfile.write_text("10.0.0.5 30050 j-worker-0 slots=1\n")
+    r = subprocess.run(
+        [sys.executable, "-m", "dgl_operator_amd.operator_plane.watcher",
+         "--watcherfile", str(watchfile), "--mode", "ready",
+         "--timeout", "10"],
+        capture_output=True, text=True, cwd="/root/repo",
+        env={**__import__('os').environ, "PATH": f"{tmp_path}:" + __import__('os').environ["PATH"]},
+    )
+    assert r.returncode == 0, r.stderr
+    assert "waiting for 1 pods" in r.stdout
