@@ -15,9 +15,12 @@ Run (driver contract):
       --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
 
 One process per GPU over RCCL; each rank owns one range partition
-(weak scaling: per-rank batch fixed at 1000 seeds). The timed step includes
-GPU neighbor sampling, alltoallv halo feature pull, forward, backward,
-gradient all-reduce and the Adam update. Metric value = SUM over ranks of
+(weak scaling: per-rank batch fixed at 1000 seeds). Multi-rank runs
+replicate the ghost-zone halo once at setup (--no-halo for per-step
+alltoallv pulls instead). The timed step includes GPU neighbor sampling,
+the feature gather, forward, backward, gradient all-reduce and the Adam
+update; 1-GPU runs replay the whole step as a hipGraph when capture
+succeeds (--no-capture for eager). Metric value = SUM over ranks of
 message-passing edges in the sampled blocks / elapsed (max over ranks).
 """
 from __future__ import annotations
