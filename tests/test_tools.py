@@ -129,3 +129,53 @@ def test_partition_balance_objectives(tmp_path):
     assert min(tloads) > 0
     assert max(tloads) <= 2.5 * max(1, min(tloads)), tloads
     assert max(eloads) <= 2.5 * max(1, min(eloads)), eloads
+
+
+def test_bench_deterministic_on_cpu(tmp_path):
+    """Same seeds -> identical edge totals (samplers are counter-based)."""
+    import json
+    import subprocess
+    import sys
+
+    REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    vals = []
+    for _ in range(2):
+        r = subprocess.run(
+            [sys.executable, "bench.py", "--steps", "2", "--warmup", "0",
+             "--nodes", "5000", "--edges", "30000"],
+            capture_output=True, text=True, cwd=REPO, timeout=240,
+        )
+        assert r.returncode == 0, r.stderr
+        d = json.loads(
+            [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+        )
+        vals.append(d["value"] * d["ms_per_step"])  # edges per step
+    assert abs(vals[0] - vals[1]) / vals[0] < 1e-6
+
+
+def test_kubexec_fabric_command_contract(monkeypatch):
+    """KubexecFabric shells through the operator's kubexec.sh / kubectl paths
+    (the ConfigMap contract, dgljob_controller.go:874-893)."""
+    import subprocess as sp
+
+    from dgl_operator_amd.tools.fabric import KubexecFabric
+
+    calls = []
+
+    class FakeProc:
+        def wait(self):
+            return 0
+
+    def fake_popen(cmd, shell=None):
+        calls.append(cmd)
+        return FakeProc()
+
+    monkeypatch.setattr(sp, "Popen", fake_popen)
+    monkeypatch.setattr(sp, "call", lambda cmd, shell=None: calls.append(cmd) or 0)
+    fab = KubexecFabric(kubexec_path="/etc/dgl/kubexec.sh",
+                        kubectl_path="/opt/kube/kubectl")
+    fab.exec("job-worker-0", "echo hi", env={"A": "1"})
+    fab.copy("/tmp/x", "job-launcher", "/w/x", container="watcher-loop-partitioner")
+    assert calls[0].startswith("/etc/dgl/kubexec.sh job-worker-0 A=1 echo hi")
+    assert calls[1] == ("/opt/kube/kubectl cp /tmp/x job-launcher:/w/x"
+                       " -c watcher-loop-partitioner")
