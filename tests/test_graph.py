@@ -77,3 +77,26 @@ def test_block_csr_transpose():
     rindptr, rindices, reids = b.csr()
     assert rindptr.tolist() == [0, 1, 2, 3, 4, 5]
     assert rindices.tolist() == [0, 1, 2, 0, 1]
+
+
+def test_rmat_chunked_generation_deterministic():
+    from dgl_operator_amd.graph.rmat import rmat_edges
+
+    s1, d1 = rmat_edges(500, 5000, seed=3, chunk=1 << 10)
+    s2, d2 = rmat_edges(500, 5000, seed=3, chunk=1 << 10)
+    assert torch.equal(s1, s2) and torch.equal(d1, d2)
+    assert s1.numel() > 4000
+    assert int(s1.max()) < 500 and int(d1.max()) < 500
+
+
+def test_block_device_roundtrip():
+    indptr = torch.tensor([0, 2, 3])
+    indices = torch.tensor([0, 2, 1])
+    b = Block(indptr, indices, num_src=3, num_dst=2,
+              srcdata_nids=torch.tensor([5, 7, 9]))
+    b.edata["w"] = torch.rand(3)
+    b2 = b.to("cpu")
+    assert torch.equal(b2.csc_indptr, indptr)
+    assert torch.equal(b2.srcdata_nids, b.srcdata_nids)
+    assert torch.equal(b2.edata["w"], b.edata["w"])
+    assert b2.num_src_nodes == 3 and b2.num_dst_nodes == 2
