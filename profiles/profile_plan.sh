@@ -34,4 +34,8 @@ rocprofv3 --kernel-trace --stats -d "$OUT/ke_trace" -- \
     --log-interval 25 --num-entities 1000000 --num-relations 1000 \
     --num-triples 2000000 > "$OUT/ke_trace.log" 2>&1
 
+# 4. per-kernel A/B microbenchmarks (HIP vs eager torch compositions)
+timeout 600 python profiles/kernel_bench.py --iters 100 \
+    > "$OUT/kernel_bench.log" 2>&1
+
 ls -R "$OUT"
