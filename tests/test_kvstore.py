@@ -191,3 +191,17 @@ def test_export_ke_npy(tmp_path):
     assert merged.shape == (10, 4)
     e0 = ShardedEmbedding(10, 4, 2, 0, seed=5)
     assert np.allclose(merged[:5], e0.local.numpy())
+
+
+def test_kge_eval_corrupt_head():
+    from dgl_operator_amd.distributed import DistKGEModel
+    from dgl_operator_amd.distributed.kge import evaluate_kge
+
+    torch.manual_seed(2)
+    m = DistKGEModel(50, 4, 8, score_func="TransE_l2", rank=0, world_size=1)
+    h = torch.randint(0, 50, (40,))
+    r = torch.randint(0, 4, (40,))
+    t = torch.randint(0, 50, (40,))
+    res = evaluate_kge(m, h, r, t, corrupt="head")
+    assert 0.0 < res["MRR"] <= 1.0
+    assert 1.0 <= res["MR"] <= 50.0
