@@ -116,3 +116,45 @@ def test_reorder_segments_roundtrip(counts_list, seed):
     for o in range(n):
         seg = out[int(off[o]) : int(off[o + 1])]
         assert bool((seg == o).all())
+
+
+@settings(max_examples=25, deadline=None)
+@given(edges_strategy, st.integers(0, 50))
+def test_builtin_equals_udf_everywhere(edges, seed):
+    """fn.u_mul_e+mean == the UDF pair on arbitrary multigraphs."""
+    import dgl_operator_amd.fn as fn
+
+    gen = torch.Generator().manual_seed(seed)
+    src = torch.tensor([e[0] for e in edges])
+    dst = torch.tensor([e[1] for e in edges])
+    g = Graph(src, dst, 20)
+    g.ndata["h"] = torch.randn(20, 3, generator=gen)
+    g.edata["w"] = torch.rand(len(edges), generator=gen)
+    g.update_all(fn.u_mul_e("h", "w", "m"), fn.mean("m", "hb"))
+
+    def msg(e):
+        return {"m": e.src["h"] * e.data["w"].unsqueeze(-1)}
+
+    def red(n):
+        return {"hu": n.mailbox["m"].mean(1)}
+
+    g.update_all(msg, red)
+    assert torch.allclose(g.ndata["hb"], g.ndata["hu"], atol=1e-5)
+
+
+def test_multihead_weight_fn_api():
+    import dgl_operator_amd.fn as fn
+    from dgl_operator_amd.graph import rmat_graph
+
+    g = rmat_graph(30, 200, seed=1)
+    H, D = 2, 4
+    g.ndata["z"] = torch.randn(30, H, D)
+    g.edata["a"] = torch.rand(g.num_edges, H)
+    g.update_all(fn.u_mul_e("z", "a", "m"), fn.sum("m", "out"))
+    assert g.ndata["out"].shape == (30, H, D)
+    # reference
+    from dgl_operator_amd.ops.spmm import _spmm_ref
+
+    indptr, indices, eids = g.csc()
+    ref = _spmm_ref(indptr, indices, g.ndata["z"], g.edata["a"][eids], False)
+    assert torch.allclose(g.ndata["out"], ref, atol=1e-5)
