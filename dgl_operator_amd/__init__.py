@@ -33,5 +33,7 @@ __version__ = "0.1.0"
 
 try:  # compute plane needs torch; the control-plane-only manager image
     from . import graph  # noqa: F401
+    from . import fn  # noqa: F401
+    from .graph import Graph, Block, batch_graphs  # noqa: F401
 except ImportError:  # pragma: no cover
     graph = None

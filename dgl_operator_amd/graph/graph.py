@@ -141,6 +141,10 @@ class Graph:
         g.ndata = dict(self.ndata)
         return g
 
+    def __repr__(self) -> str:
+        return (f"Graph(num_nodes={self._num_nodes}, "
+                f"num_edges={self.num_edges}, device={self.device})")
+
     # -- DGL-style message passing API ------------------------------------
     def update_all(self, message_func, reduce_func) -> None:
         """g.update_all(fn.copy_u('h','m'), fn.mean('m','h_N')) — builtin
@@ -294,6 +298,10 @@ class Block:
 
     def in_degrees(self) -> torch.Tensor:
         return self.csc_indptr[1:] - self.csc_indptr[:-1]
+
+    def __repr__(self) -> str:
+        return (f"Block(num_src={self._num_src}, num_dst={self._num_dst}, "
+                f"num_edges={self.num_edges}, device={self.device})")
 
     def update_all(self, message_func, reduce_func) -> None:
         """Blocks support the fn API too; reduced fields land in
