@@ -70,6 +70,7 @@ class SAGEConv(nn.Module):
                 x.feat.is_cuda
                 and x.feat.dtype == torch.float32
                 and self.fc_neigh.out_features <= 16
+                and x.feat.shape[1] <= 230  # gather_mm LDS envelope
             )
             if not fusable:
                 x = x.materialize()
