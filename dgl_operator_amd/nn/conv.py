@@ -148,7 +148,9 @@ class GATConv(nn.Module):
     """
 
     def __init__(self, in_feats: int, out_feats: int, num_heads: int = 1,
-                 negative_slope: float = 0.2, bias: bool = True):
+                 negative_slope: float = 0.2, bias: bool = True,
+                 feat_drop: float = 0.0, attn_drop: float = 0.0,
+                 residual: bool = False):
         super().__init__()
         self.num_heads = num_heads
         self.out_feats = out_feats
@@ -157,6 +159,14 @@ class GATConv(nn.Module):
         self.attn_r = nn.Parameter(torch.empty(num_heads, out_feats))
         self.bias = nn.Parameter(torch.zeros(num_heads * out_feats)) if bias else None
         self.leaky = nn.LeakyReLU(negative_slope)
+        self.feat_drop = nn.Dropout(feat_drop)
+        self.attn_drop = nn.Dropout(attn_drop)
+        self.res_fc = None
+        if residual:
+            self.res_fc = (
+                nn.Identity() if in_feats == out_feats * num_heads
+                else nn.Linear(in_feats, out_feats * num_heads, bias=False)
+            )
         self.reset_parameters()
 
     def reset_parameters(self):
@@ -167,14 +177,17 @@ class GATConv(nn.Module):
 
     def forward(self, g: Union[Graph, Block], x: torch.Tensor) -> torch.Tensor:
         H, D = self.num_heads, self.out_feats
+        x = self.feat_drop(x)
         z = self.fc(x).view(-1, H, D)  # [N, H, D]
         el = (z * self.attn_l).sum(-1)  # [N, H]
         er = (z * self.attn_r).sum(-1)
         # fused u_add_v + LeakyReLU attention logits (csc order)
         score = gat_score(g, el, er, self.leaky.negative_slope)
-        alpha = edge_softmax_csc(g, score)
+        alpha = self.attn_drop(edge_softmax_csc(g, score))
         out = gspmm(g, "u_mul_e", "sum", z, _csc_weight(g, alpha))  # [Nd, H, D]
         out = out.reshape(-1, H * D)
+        if self.res_fc is not None:
+            out = out + self.res_fc(_dst_feat(g, x))
         if self.bias is not None:
             out = out + self.bias
         return out

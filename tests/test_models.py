@@ -161,3 +161,22 @@ def test_graphconv_matches_dense_normalization():
     dout = A.sum(0).clamp(min=1)
     ref = torch.diag(din.pow(-0.5)) @ A @ torch.diag(dout.pow(-0.5)) @ x @ layer.weight
     assert torch.allclose(out, ref, atol=1e-4)
+
+
+def test_gatconv_options():
+    from dgl_operator_amd.nn import GATConv
+
+    g = rmat_graph(30, 200, num_feats=12, seed=4)
+    layer = GATConv(12, 6, num_heads=2, feat_drop=0.1, attn_drop=0.1,
+                    residual=True)
+    layer.train()
+    out = layer(g, g.ndata["feat"])
+    assert out.shape == (30, 12)
+    out.sum().backward()
+    layer.eval()
+    o1 = layer(g, g.ndata["feat"])
+    o2 = layer(g, g.ndata["feat"])
+    assert torch.allclose(o1, o2)  # dropout off in eval
+    # residual identity when dims match
+    layer2 = GATConv(12, 12, num_heads=1, residual=True)
+    assert isinstance(layer2.res_fc, torch.nn.Identity)
