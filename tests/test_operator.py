@@ -362,3 +362,28 @@ def test_kubectl_cluster_and_watcher_cli(tmp_path, monkeypatch):
     )
     assert r.returncode == 0, r.stderr
     assert "waiting for 1 pods" in r.stdout
+
+
+def test_manifest_roundtrip_serialization():
+    import glob
+    import os
+
+    import yaml as _yaml
+
+    from dgl_operator_amd.operator_plane import job_to_manifest
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    for path in glob.glob(os.path.join(repo, "examples/v1alpha1/*.yaml")):
+        with open(path) as f:
+            raw = _yaml.safe_load(f)
+        job = job_from_manifest(raw)
+        out = job_to_manifest(job)
+        job2 = job_from_manifest(out)
+        assert job2.name == job.name
+        assert job2.spec.partition_mode == job.spec.partition_mode
+        assert job2.spec.clean_pod_policy == job.spec.clean_pod_policy
+        assert job2.spec.slots_per_worker == job.spec.slots_per_worker
+        assert set(job2.spec.replica_specs) == set(job.spec.replica_specs)
+        for rt, rs in job.spec.replica_specs.items():
+            assert job2.spec.replica_specs[rt].replicas == rs.replicas
+            assert job2.spec.replica_specs[rt].template == rs.template
