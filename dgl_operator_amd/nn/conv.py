@@ -40,15 +40,19 @@ class SAGEConv(nn.Module):
     def __init__(self, in_feats: int, out_feats: int, aggregator: str = "mean",
                  bias: bool = True):
         super().__init__()
-        assert aggregator in ("mean", "sum"), aggregator
+        assert aggregator in ("mean", "sum", "gcn"), aggregator
         self.aggregator = aggregator
-        self.fc_self = nn.Linear(in_feats, out_feats, bias=False)
+        # 'gcn' folds the self feature into the normalized aggregate (DGL
+        # SAGEConv aggregator_type='gcn'): no separate fc_self
+        self.fc_self = (None if aggregator == "gcn"
+                        else nn.Linear(in_feats, out_feats, bias=False))
         self.fc_neigh = nn.Linear(in_feats, out_feats, bias=bias)
         self.reset_parameters()
 
     def reset_parameters(self):
         gain = nn.init.calculate_gain("relu")
-        nn.init.xavier_uniform_(self.fc_self.weight, gain=gain)
+        if self.fc_self is not None:
+            nn.init.xavier_uniform_(self.fc_self.weight, gain=gain)
         nn.init.xavier_uniform_(self.fc_neigh.weight, gain=gain)
         if self.fc_neigh.bias is not None:
             nn.init.zeros_(self.fc_neigh.bias)
@@ -65,6 +69,16 @@ class SAGEConv(nn.Module):
         less HBM traffic at the bench shape 100 -> 16). Zero-degree rows
         match too: both orders yield exactly the bias."""
         op = "copy_u" if edge_weight is None else "u_mul_e"
+        if self.aggregator == "gcn":
+            if isinstance(x, GatherView):
+                x = x.materialize()
+            agg = gspmm(g, op, "sum", x, edge_weight)
+            xd = _dst_feat(g, x)
+            deg = (
+                g.in_degrees() if not isinstance(g, Block)
+                else (g.csc_indptr[1:] - g.csc_indptr[:-1])
+            ).to(x.dtype).unsqueeze(-1)
+            return self.fc_neigh((agg + xd) / (deg + 1))
         if isinstance(x, GatherView):
             fusable = (
                 x.feat.is_cuda

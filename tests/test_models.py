@@ -180,3 +180,21 @@ def test_gatconv_options():
     # residual identity when dims match
     layer2 = GATConv(12, 12, num_heads=1, residual=True)
     assert isinstance(layer2.res_fc, torch.nn.Identity)
+
+
+def test_sageconv_gcn_aggregator():
+    from dgl_operator_amd.nn import SAGEConv
+
+    g = rmat_graph(25, 150, num_feats=6, seed=2)
+    layer = SAGEConv(6, 4, aggregator="gcn")
+    x = g.ndata["feat"]
+    out = layer(g, x)
+    assert out.shape == (25, 4)
+    # manual reference: fc((sum_neighbors + self) / (deg + 1))
+    from dgl_operator_amd.ops import gspmm
+
+    agg = gspmm(g, "copy_u", "sum", x)
+    deg = g.in_degrees().to(x.dtype).unsqueeze(-1)
+    ref = layer.fc_neigh((agg + x) / (deg + 1))
+    assert torch.allclose(out, ref, atol=1e-5)
+    out.sum().backward()
