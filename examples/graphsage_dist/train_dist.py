@@ -117,9 +117,19 @@ def main():
     if args.checkpoint_path:
         os.makedirs(args.checkpoint_path, exist_ok=True)
         ckpt_file = os.path.join(args.checkpoint_path, "graphsage.pt")
-        if os.path.exists(ckpt_file):
-            state = torch.load(ckpt_file, map_location=device,
+        # rank 0 saves; on resume the state is BROADCAST so ranks without a
+        # shared checkpoint volume (pod-local dirs) stay consistent (epoch
+        # counter AND Adam moments — divergent moments would desync params
+        # despite the gradient all-reduce)
+        state = None
+        if rank == 0 and os.path.exists(ckpt_file):
+            state = torch.load(ckpt_file, map_location="cpu",
                                weights_only=True)
+        if ws > 1:
+            box = [state]
+            dist.broadcast_object_list(box, src=0)
+            state = box[0]
+        if state is not None:
             model.load_state_dict(state["model"])
             opt.load_state_dict(state["optimizer"])
             start_epoch = state["epoch"] + 1

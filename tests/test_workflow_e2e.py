@@ -109,3 +109,40 @@ def test_train_dist_single_node_multi_rank(tmp_path):
             break
     assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
     assert "Eval acc" in r.stdout
+
+
+@pytest.mark.timeout(500)
+def test_train_dist_checkpoint_resume(tmp_path):
+    """2-rank resume with a rank0-only (pod-local) checkpoint: state is
+    broadcast so all ranks restart from the same epoch + optimizer state."""
+    r = subprocess.run(
+        [sys.executable,
+         os.path.join(REPO, "examples/graphsage_dist/load_and_partition_graph.py"),
+         "--graph-name", "t", "--num-partitions", "2",
+         "--output", str(tmp_path), "--nodes", "200", "--edges", "1500",
+         "--feat", "6", "--classes", "3", "--algorithm", "range"],
+        capture_output=True, text=True, cwd=REPO)
+    assert r.returncode == 0, r.stderr
+    ck = tmp_path / "ck"
+
+    def run_once(epochs):
+        for attempt in range(2):
+            r = subprocess.run(
+                [sys.executable, "-m", "torch.distributed.run", "--standalone",
+                 "--local-addr", "127.0.0.1", "--nproc-per-node", "2",
+                 os.path.join(REPO, "examples/graphsage_dist/train_dist.py"),
+                 "--part-config", str(tmp_path / "t.json"),
+                 "--num-epochs", str(epochs), "--batch-size", "16",
+                 "--fan-out", "2,2", "--log-every", "100",
+                 "--checkpoint-path", str(ck)],
+                capture_output=True, text=True, cwd=REPO, timeout=240)
+            if r.returncode == 0:
+                return r
+        assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+
+    run_once(1)
+    assert (ck / "graphsage.pt").exists()
+    r2 = run_once(2)
+    assert "resumed from epoch 0" in r2.stdout
+    assert "Epoch 001 time" in r2.stdout
+    assert "Epoch 000 time" not in r2.stdout  # skipped the finished epoch
