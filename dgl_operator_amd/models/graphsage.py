@@ -64,10 +64,24 @@ def inference_dist(model: GraphSAGE, dg, batch_size: int = 1000,
     cur_key = feat_key
     owned = dg.owned_nodes()
     n_layers = len(model.layers)
+    # every rank must issue the SAME number of pull collectives per layer:
+    # align batch counts to the global max (ranks that run out issue empty
+    # batches, which still participate in the exchange)
+    import math
+
+    n_batches = max(1, math.ceil(owned.numel() / batch_size))
+    from ..distributed import comm as _comm
+
+    if _comm.world()[1] > 1:
+        import torch.distributed as dist
+
+        nb = torch.tensor([n_batches])
+        dist.all_reduce(nb, op=dist.ReduceOp.MAX)
+        n_batches = int(nb[0])
     for li, layer in enumerate(model.layers):
         outs = []
-        for s in range(0, owned.numel(), batch_size):
-            seeds = owned[s : s + batch_size]
+        for bi in range(n_batches):
+            seeds = owned[bi * batch_size : (bi + 1) * batch_size]
             blk = dg.full_neighbor_block(seeds)
             x = dg.pull(cur_key, blk.srcdata_nids)
             h = layer(blk, x)

@@ -140,7 +140,13 @@ def main():
 
     gen = torch.Generator(device=device)
     gen.manual_seed(1234 + rank)
+    # ranks own different train-node counts; every rank must run the SAME
+    # number of steps or the gradient all-reduce deadlocks — reduce to MIN
     steps_per_epoch = max(1, train_nids.numel() // args.batch_size)
+    if ws > 1:
+        spe = torch.tensor([steps_per_epoch])
+        dist.all_reduce(spe, op=dist.ReduceOp.MIN)
+        steps_per_epoch = max(1, int(spe[0]))
 
     for epoch in range(start_epoch, args.num_epochs):
         t_epoch = time.time()

@@ -291,3 +291,31 @@ def _torch_ddp_worker(rank, world):
 
 def test_torch_ddp_compat():
     _run_workers(_torch_ddp_worker)
+
+
+def _uneven_worker(rank, world):
+    """Uneven shards (rank0 owns 3x the nodes): inference and training must
+    not deadlock on mismatched collective counts."""
+    import torch.nn.functional as F
+
+    from dgl_operator_amd.distributed import DistGraph
+    from dgl_operator_amd.graph import rmat_graph
+    from dgl_operator_amd.models import GraphSAGE
+    from dgl_operator_amd.models.graphsage import inference_dist
+
+    g = rmat_graph(200, 2500, num_feats=6, num_classes=3, seed=4)
+    book = PartitionBook([0, 150, 200])  # 150 vs 50 owned
+    dg = DistGraph.from_full_graph(g, book, rank)
+    model = GraphSAGE(6, 8, 3, n_layers=2, dropout=0.0)
+    for p in model.parameters():
+        dist.broadcast(p.data, src=0)
+    # batch smaller than the big shard but bigger than small-shard batches
+    shard = inference_dist(model, dg, batch_size=60)
+    model.eval()
+    with torch.no_grad():
+        full = model(g, g.ndata["feat"])
+    assert torch.allclose(shard, full[dg.lo : dg.hi], atol=1e-4, rtol=1e-4)
+
+
+def test_uneven_shards_no_deadlock():
+    _run_workers(_uneven_worker)
