@@ -48,7 +48,7 @@ def _capture_ke_step(model, sampler, args, device):
         rc = r.view(C, args.chunk_size, -1)
         nc = n.reshape(C, args.neg_sample_size, -1)
         neg = model.score.neg(hc, rc, nc, neg_head=neg_head)
-        loss = kge_loss(pos, neg, 1.0)
+        loss = kge_loss(pos, neg, args.adversarial_temperature)
         loss.backward()
         with th.no_grad():
             model.entities.push_grad(ent_ids, rows.grad, args.lr)
@@ -102,6 +102,9 @@ def main():
     p.add_argument("--batch-size", type=int, default=1024)
     p.add_argument("--neg-sample-size", type=int, default=256)
     p.add_argument("--chunk-size", type=int, default=64)
+    p.add_argument("--adversarial-temperature", "-adv-temp", type=float,
+                   default=1.0, help="self-adversarial sampling temperature "
+                                     "(dglke -adv is always on here)")
     p.add_argument("--max-step", type=int, default=1000)
     p.add_argument("--log-interval", type=int, default=100)
     p.add_argument("--num-entities", type=int, default=1_000_000)
@@ -169,8 +172,11 @@ def main():
             loss = None
         else:
             hh, rr, tt, negs, neg_head = sampler.next_batch()
-            loss = model.train_step(hh, rr, tt, negs, args.chunk_size,
-                                    args.lr, neg_head=neg_head)
+            loss = model.train_step(
+                hh, rr, tt, negs, args.chunk_size, args.lr,
+                neg_head=neg_head,
+                adversarial_temperature=args.adversarial_temperature,
+            )
         if step % args.log_interval == 0 and rank == 0:
             if loss is None:
                 loss = float(loss_t.detach().cpu())
