@@ -33,7 +33,11 @@ def update_all_udf(gstruct, ndata: Dict[str, torch.Tensor],
                    edata: Dict[str, torch.Tensor],
                    message_func: Callable, reduce_func: Callable,
                    num_dst: int):
-    """Returns a dict of reduced node fields over in-edges of every dst."""
+    """Returns a dict of reduced node fields over in-edges of every dst.
+
+    Destinations with no in-edges get zero rows (DGL semantics); on a graph
+    with NO edges at all the reduce never runs, so no output fields are
+    produced (output shapes are only known from the reducer)."""
     indptr, indices, eids = gstruct.csc()
     from .spmm import _edge_dst
 
