@@ -43,6 +43,9 @@ def test_bench_json_contract():
     ("examples/node_classification/train.py", ["--epochs", "3", "--feat", "32"]),
     ("examples/link_predict/train.py",
      ["--epochs", "3", "--nodes", "300", "--edges", "2000"]),
+    ("examples/link_predict/train.py",
+     ["--minibatch", "--epochs", "1", "--nodes", "500", "--edges", "4000",
+      "--batch-edges", "64", "--steps", "4"]),
     ("examples/graph_classification/train.py",
      ["--epochs", "2", "--num-graphs", "20"]),
     ("examples/message_passing/train.py",
