@@ -28,7 +28,7 @@ def test_graphsage_minibatch_trains():
         opt.zero_grad()
         loss.backward()
         opt.step()
-        losses.append(float(loss))
+        losses.append(float(loss.detach()))
     assert losses[-1] < losses[0]
 
 
