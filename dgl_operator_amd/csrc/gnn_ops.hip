@@ -158,20 +158,28 @@ static void spmm_launch(const at::Tensor& indptr, const at::Tensor& indices,
   const int64_t thresh =
       long_ok ? kSpmmLongRow : std::numeric_limits<int64_t>::max();
   auto stream = cur_stream();
+// NOTE: must be a single statement (do/while) — an unbraced two-statement
+// expansion under `if (wm==…) DOA_SPMM(…); else …` re-binds the else to the
+// inner `if (long_ok)` and launches the W_SCALAR/W_HEAD long kernels with a
+// null edge-weight pointer (GPU memory fault on any >thresh-degree row).
 #define DOA_SPMM(V, W)                                                        \
-  hipLaunchKernelGGL((spmm_kernel<scalar_t, V, W>), dim3(grid), dim3(block),  \
-                     0, stream, indptr.data_ptr<int64_t>(),                   \
-                     indices.data_ptr<int64_t>(), feat.data_ptr<scalar_t>(),  \
-                     ewp, out.data_ptr<scalar_t>(), num_rows, F, D, mean,     \
-                     thresh);                                                 \
-  if (long_ok)                                                                \
-    hipLaunchKernelGGL((spmm_long_kernel<scalar_t, V, W>),                    \
-                       dim3(grid_for(num_rows * 256, block)), dim3(block),    \
-                       F * sizeof(float), stream, indptr.data_ptr<int64_t>(), \
+  do {                                                                        \
+    hipLaunchKernelGGL((spmm_kernel<scalar_t, V, W>), dim3(grid),             \
+                       dim3(block), 0, stream, indptr.data_ptr<int64_t>(),    \
                        indices.data_ptr<int64_t>(),                           \
                        feat.data_ptr<scalar_t>(), ewp,                        \
                        out.data_ptr<scalar_t>(), num_rows, F, D, mean,        \
-                       kSpmmLongRow)
+                       thresh);                                               \
+    if (long_ok)                                                              \
+      hipLaunchKernelGGL((spmm_long_kernel<scalar_t, V, W>),                  \
+                         dim3(grid_for(num_rows * 256, block)), dim3(block),  \
+                         F * sizeof(float), stream,                           \
+                         indptr.data_ptr<int64_t>(),                          \
+                         indices.data_ptr<int64_t>(),                         \
+                         feat.data_ptr<scalar_t>(), ewp,                      \
+                         out.data_ptr<scalar_t>(), num_rows, F, D, mean,      \
+                         kSpmmLongRow);                                       \
+  } while (0)
   if (vec4) {
     if (wm == W_NONE) DOA_SPMM(4, W_NONE);
     else if (wm == W_SCALAR) DOA_SPMM(4, W_SCALAR);

@@ -156,10 +156,34 @@ def evaluate_kge(
     (h, r, t') with t' != t are excluded from the ranking by subtracting the
     known competitors that outscore the true entity (every rank is assumed
     to hold the full triple list, as in this repo's synthetic KGs).
+
+    Collective contract: every ``pull`` below is an alltoallv, so ALL RANKS
+    MUST PASS IDENTICAL (heads, rels, tails) AND filter sets — asserted up
+    front. The filtered rescoring issues exactly ONE batched pull per eval
+    batch (never a data-dependent number of collectives per triple).
+
+    Tie handling is optimistic ('>'): the true entity itself sits in the
+    candidate shard, and its score there comes from the chunked ``neg``
+    kernel while ``true_score`` comes from ``edge`` — bitwise equality
+    between the two paths is not guaranteed, so a '>=' (pessimistic) count
+    could penalize the truth against itself.
     """
     import torch.distributed as dist
 
     device = heads.device
+    if dist.is_available() and dist.is_initialized() and \
+            dist.get_world_size() > 1:
+        dev = model.entities.local.device \
+            if dist.get_backend() == "nccl" else torch.device("cpu")
+        sig = torch.tensor(
+            [float(heads.numel()), float(heads.sum()), float(rels.sum()),
+             float(tails.sum())], dtype=torch.float64, device=dev)
+        lo, hi = sig.clone(), sig.clone()
+        dist.all_reduce(lo, op=dist.ReduceOp.MIN)
+        dist.all_reduce(hi, op=dist.ReduceOp.MAX)
+        assert torch.equal(lo, hi), (
+            "evaluate_kge: all ranks must evaluate the SAME triples "
+            "(every pull is a collective; divergent loops deadlock)")
     rank_sum = 0.0
     rr_sum = 0.0
     hit_counts = {k: 0.0 for k in hits}
@@ -201,25 +225,35 @@ def evaluate_kge(
         if dist.is_available() and dist.is_initialized():
             dist.all_reduce(higher)
         if known is not None:
-            # filtered: remove known competitors that outscored the truth
+            # filtered: remove known competitors that outscored the truth.
+            # ONE batched pull for the whole eval batch (identical on all
+            # ranks by the contract asserted above), then local rescoring.
             adj = torch.zeros_like(higher)
+            cand_lists = []
             for i in range(B):
                 key = (int(hh[i]), int(rr[i]))
-                cand_ids = [e for e in known.get(key, ())
-                            if e != int(tt[i] if corrupt == "tail" else hh[i])]
-                if not cand_ids:
-                    continue
-                ids_t = torch.as_tensor(cand_ids, device=device)
-                ke = model.entities.pull(ids_t)
-                if corrupt == "tail":
-                    sc = model.score.edge(
-                        h[i].expand(len(cand_ids), -1),
-                        r[i].expand(len(cand_ids), -1), ke)
-                else:
-                    sc = model.score.edge(
-                        ke, r[i].expand(len(cand_ids), -1),
-                        t[i].expand(len(cand_ids), -1))
-                adj[i] = (sc > true_score[i]).sum()
+                excl = int(tt[i] if corrupt == "tail" else hh[i])
+                cand_lists.append(
+                    [e for e in known.get(key, ()) if e != excl])
+            flat = [e for lst in cand_lists for e in lst]
+            if flat:
+                ke_all = model.entities.pull(
+                    torch.as_tensor(flat, device=device))
+                off = 0
+                for i, lst in enumerate(cand_lists):
+                    if not lst:
+                        continue
+                    ke = ke_all[off : off + len(lst)]
+                    off += len(lst)
+                    if corrupt == "tail":
+                        sc = model.score.edge(
+                            h[i].expand(len(lst), -1),
+                            r[i].expand(len(lst), -1), ke)
+                    else:
+                        sc = model.score.edge(
+                            ke, r[i].expand(len(lst), -1),
+                            t[i].expand(len(lst), -1))
+                    adj[i] = (sc > true_score[i]).sum()
             higher = (higher - adj).clamp(min=0)
         ranks = higher + 1.0
         rank_sum += float(ranks.sum())

@@ -179,7 +179,7 @@ def _update_all(g, message_func, reduce_func, num_dst: int) -> None:
         )
         return
     out = update_all_udf(g, src_fields, g.edata, message_func, reduce_func,
-                         num_dst)
+                         num_dst, dstdata=g.dstdata if is_block else None)
     out_fields.update(out)
 
 
@@ -323,6 +323,8 @@ class Block:
         if self._csr is not None:
             b._csr = tuple(t.to(device) for t in self._csr)
         b.edata = {k: v.to(device) for k, v in self.edata.items()}
+        b.srcdata = {k: v.to(device) for k, v in self.srcdata.items()}
+        b.dstdata = {k: v.to(device) for k, v in self.dstdata.items()}
         return b
 
 

@@ -75,7 +75,10 @@ def inference_dist(model: GraphSAGE, dg, batch_size: int = 1000,
     if _comm.world()[1] > 1:
         import torch.distributed as dist
 
-        nb = torch.tensor([n_batches])
+        # RCCL (backend 'nccl') rejects CPU tensors — reduce on the shard's
+        # compute device (gloo accepts either)
+        nb_dev = owned.device if dist.get_backend() == "nccl" else "cpu"
+        nb = torch.tensor([n_batches], device=nb_dev)
         dist.all_reduce(nb, op=dist.ReduceOp.MAX)
         n_batches = int(nb[0])
     for li, layer in enumerate(model.layers):

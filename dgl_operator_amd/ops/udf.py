@@ -32,18 +32,28 @@ class NodeBatch:
 def update_all_udf(gstruct, ndata: Dict[str, torch.Tensor],
                    edata: Dict[str, torch.Tensor],
                    message_func: Callable, reduce_func: Callable,
-                   num_dst: int):
+                   num_dst: int, dstdata: Dict[str, torch.Tensor] = None):
     """Returns a dict of reduced node fields over in-edges of every dst.
 
     Destinations with no in-edges get zero rows (DGL semantics); on a graph
     with NO edges at all the reduce never runs, so no output fields are
-    produced (output shapes are only known from the reducer)."""
+    produced (output shapes are only known from the reducer).
+
+    For Blocks ``dstdata`` carries fields stored per-dst-row; dst-side
+    lookups (``edges.dst[...]``, ``nodes.data[...]``) read it with
+    precedence over ``ndata`` — relying on the dst-first src-row convention
+    alone would make dst-only fields invisible to UDFs."""
     indptr, indices, eids = gstruct.csc()
     from .spmm import _edge_dst
 
     dst = _edge_dst(indptr)
     src_data = {k: v[indices] for k, v in ndata.items()}
-    dst_data = {k: v[dst] for k, v in ndata.items()}
+    # dst-side view: dst rows are the first num_dst src rows by convention,
+    # but fields assigned only on dstdata must win / still be visible
+    dst_side = dict(ndata)
+    if dstdata:
+        dst_side.update(dstdata)
+    dst_data = {k: v[dst] for k, v in dst_side.items()}
     if eids is not None:
         edge_data = {k: v[eids] for k, v in edata.items()}
     else:
@@ -62,7 +72,7 @@ def update_all_udf(gstruct, ndata: Dict[str, torch.Tensor],
         mailbox = {k: v[pos.reshape(-1)].view(rows.numel(), d, *v.shape[1:])
                    for k, v in msgs.items()}
         reduced = reduce_func(
-            NodeBatch({k: v[rows] for k, v in ndata.items()}, mailbox)
+            NodeBatch({k: v[rows] for k, v in dst_side.items()}, mailbox)
         )
         for k, v in reduced.items():
             if k not in out:

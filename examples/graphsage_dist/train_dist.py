@@ -144,7 +144,8 @@ def main():
     # number of steps or the gradient all-reduce deadlocks — reduce to MIN
     steps_per_epoch = max(1, train_nids.numel() // args.batch_size)
     if ws > 1:
-        spe = torch.tensor([steps_per_epoch])
+        # RCCL (backend 'nccl') rejects CPU tensors — reduce on the device
+        spe = torch.tensor([steps_per_epoch], device=device)
         dist.all_reduce(spe, op=dist.ReduceOp.MIN)
         steps_per_epoch = max(1, int(spe[0]))
 
