@@ -149,9 +149,26 @@ def _build_captured_step(args, dg, model, opt, device, fanouts, next_seeds,
             fill_seeds(s)
             capture_body()
     torch.cuda.current_stream().wait_stream(side)
-    graph = torch.cuda.CUDAGraph()
-    with torch.cuda.graph(graph):
-        capture_body()
+    # capture-mode ladder: "global" errors if ANY thread touches the HIP API
+    # unsafely during capture — backward() runs on autograd worker threads,
+    # which trips hipErrorStreamCaptureUnsupported on some ROCm runtimes.
+    # thread_local/relaxed scope the restriction to the capturing thread.
+    graph = None
+    last_err = None
+    for mode in ("global", "thread_local", "relaxed"):
+        gr = torch.cuda.CUDAGraph()
+        try:
+            with torch.cuda.graph(gr, capture_error_mode=mode):
+                capture_body()
+            graph = gr
+            if rank == 0 and mode != "global":
+                print(f"# capture: mode={mode}")
+            break
+        except Exception as e:  # noqa: BLE001
+            last_err = e
+            torch.cuda.synchronize()
+    if graph is None:
+        raise last_err
     # sanity: two replays must accumulate edges and keep weights finite
     edge_accum.zero_()
     for s in range(2):
@@ -293,7 +310,8 @@ def main():
             )
         except Exception as e:  # noqa: BLE001
             use_capture = False
-            print(f"# capture: disabled ({type(e).__name__}: {e}); "
+            msg = str(e).splitlines()[0] if str(e) else type(e).__name__
+            print(f"# capture: disabled ({type(e).__name__}: {msg}); "
                   "eager stepping")
             opt = torch.optim.Adam(model.parameters(), lr=args.lr)
 

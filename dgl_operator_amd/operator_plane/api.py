@@ -150,11 +150,48 @@ def job_from_manifest(manifest: Dict[str, Any] | str) -> DGLJob:
         slots_per_worker=int(spec_d.get("slotsPerWorker", 1)),
         replica_specs=replica_specs,
     )
-    return DGLJob(
+    job = DGLJob(
         name=meta.get("name", "dgljob"),
         namespace=meta.get("namespace", "default"),
         spec=spec,
+        uid=meta.get("uid", ""),
     )
+    if meta.get("deletionTimestamp"):
+        import time as _time
+
+        job.deletion_timestamp = _time.time()
+    return job
+
+
+def _rfc3339(ts: Optional[float]) -> Optional[str]:
+    if ts is None:
+        return None
+    import datetime
+
+    return datetime.datetime.fromtimestamp(
+        ts, tz=datetime.timezone.utc).strftime("%Y-%m-%dT%H:%M:%SZ")
+
+
+def status_to_manifest(job: DGLJob) -> Dict[str, Any]:
+    """job.status -> the CR .status shape (dgljob_types.go:94-108)."""
+    st: Dict[str, Any] = {}
+    if job.status.phase is not None:
+        st["phase"] = job.status.phase.value
+    if job.status.start_time is not None:
+        st["startTime"] = _rfc3339(job.status.start_time)
+    if job.status.completion_time is not None:
+        st["completionTime"] = _rfc3339(job.status.completion_time)
+    if job.status.replica_statuses:
+        st["replicaStatuses"] = {
+            rt.value: {
+                "active": rs.active,
+                "succeeded": rs.succeeded,
+                "failed": rs.failed,
+                "ready": rs.ready,
+            }
+            for rt, rs in job.status.replica_statuses.items()
+        }
+    return st
 
 
 def job_to_manifest(job: DGLJob) -> Dict[str, Any]:

@@ -31,6 +31,7 @@ class Pod:
     ip: Optional[str] = None
     containers_ready: bool = False
     owner: Optional[str] = None  # owning DGLJob name
+    owner_uid: Optional[str] = None  # DGLJob uid (k8s ownerReference GC)
 
     def is_real_running(self) -> bool:
         """Running AND every container ready (dgljob_controller.go:1511-1528)."""
@@ -43,6 +44,7 @@ class ConfigMap:
     namespace: str
     data: Dict[str, str] = field(default_factory=dict)
     owner: Optional[str] = None
+    owner_uid: Optional[str] = None
 
 
 @dataclass
@@ -53,6 +55,7 @@ class Service:
     cluster_ip: Optional[str] = None  # None => headless
     ports: List[int] = field(default_factory=list)
     owner: Optional[str] = None
+    owner_uid: Optional[str] = None
 
 
 @dataclass
@@ -62,6 +65,7 @@ class RBACObject:
     namespace: str
     rules: List[Dict[str, Any]] = field(default_factory=list)
     owner: Optional[str] = None
+    owner_uid: Optional[str] = None
 
 
 class Cluster:
@@ -186,21 +190,47 @@ class FakeCluster(Cluster):
 
 
 class KubectlCluster(Cluster):
-    """Thin kubectl-backed Cluster for running the watcher/manager against a
-    real Kubernetes API (read paths used by watcher-loop; create/update paths
-    shell out to `kubectl apply/create`). Requires a kubeconfig."""
+    """kubectl-backed Cluster: the real-apiserver implementation of every
+    reconciler dependency. Object writes render the internal model into v1
+    manifests (k8s.py) and pipe them to ``kubectl create/apply -f -``;
+    reads parse ``-o json``. Also carries the DGLJob CR surface the Manager
+    watch loop uses (list/patch-status), replacing the reference's
+    controller-runtime informer plumbing
+    (/root/reference/main.go:73-105, dgljob_controller.go:436-458) with a
+    polling loop over the same objects. Requires a kubeconfig (in-cluster
+    service account or --kubeconfig)."""
 
     def __init__(self, kubectl: str = "kubectl"):
         self.kubectl = kubectl
 
-    def _get_json(self, args):
-        r = subprocess.run(
-            [self.kubectl] + args + ["-o", "json"],
-            capture_output=True, text=True,
+    # -- plumbing ----------------------------------------------------------
+    def _run(self, args, stdin: "str | None" = None):
+        return subprocess.run(
+            [self.kubectl] + args, input=stdin, capture_output=True,
+            text=True,
         )
+
+    def _get_json(self, args):
+        r = self._run(args + ["-o", "json"])
         if r.returncode != 0:
             return None
         return json.loads(r.stdout)
+
+    def _create(self, manifest: dict):
+        r = self._run(["create", "-f", "-"], stdin=json.dumps(manifest))
+        if r.returncode != 0 and "AlreadyExists" not in r.stderr:
+            raise RuntimeError(
+                f"kubectl create {manifest.get('kind')} "
+                f"{manifest.get('metadata', {}).get('name')}: {r.stderr.strip()}"
+            )
+
+    def _apply(self, manifest: dict):
+        r = self._run(["apply", "-f", "-"], stdin=json.dumps(manifest))
+        if r.returncode != 0:
+            raise RuntimeError(
+                f"kubectl apply {manifest.get('kind')} "
+                f"{manifest.get('metadata', {}).get('name')}: {r.stderr.strip()}"
+            )
 
     def get_pod(self, namespace, name):
         d = self._get_json(["get", "pod", name, "-n", namespace])
@@ -232,9 +262,96 @@ class KubectlCluster(Cluster):
             for item in d.get("items", [])
         ]
 
+    def create_pod(self, pod):
+        from .k8s import pod_manifest
+
+        self._create(pod_manifest(pod))
+        return pod
+
     def delete_pod(self, namespace, name):
         subprocess.run(
             [self.kubectl, "delete", "pod", name, "-n", namespace,
-             "--ignore-not-found"],
+             "--ignore-not-found", "--wait=false"],
             capture_output=True,
         )
+
+    # -- configmaps --------------------------------------------------------
+    def get_configmap(self, namespace, name):
+        d = self._get_json(["get", "configmap", name, "-n", namespace])
+        if d is None:
+            return None
+        return ConfigMap(name=name, namespace=namespace,
+                         data=d.get("data") or {})
+
+    def create_configmap(self, cm):
+        from .k8s import configmap_manifest
+
+        self._create(configmap_manifest(cm))
+        return cm
+
+    def update_configmap(self, cm):
+        from .k8s import configmap_manifest
+
+        self._apply(configmap_manifest(cm))
+        return cm
+
+    # -- services ----------------------------------------------------------
+    def get_service(self, namespace, name):
+        d = self._get_json(["get", "service", name, "-n", namespace])
+        if d is None:
+            return None
+        spec = d.get("spec", {})
+        cip = spec.get("clusterIP")
+        return Service(
+            name=name, namespace=namespace,
+            selector=spec.get("selector") or {},
+            cluster_ip=None if cip == "None" else cip,
+            ports=[p.get("port") for p in spec.get("ports", [])],
+        )
+
+    def create_service(self, svc):
+        from .k8s import service_manifest
+
+        self._create(service_manifest(svc))
+        return svc
+
+    # -- rbac ----------------------------------------------------------------
+    _RBAC_RESOURCE = {"ServiceAccount": "serviceaccount", "Role": "role",
+                      "RoleBinding": "rolebinding"}
+
+    def get_rbac(self, namespace, kind, name):
+        d = self._get_json(
+            ["get", self._RBAC_RESOURCE[kind], name, "-n", namespace])
+        if d is None:
+            return None
+        return RBACObject(kind=kind, name=name, namespace=namespace,
+                          rules=d.get("rules") or [])
+
+    def create_rbac(self, obj):
+        from .k8s import rbac_manifest
+
+        self._create(rbac_manifest(obj))
+        return obj
+
+    # -- DGLJob CRs (the Manager watch surface) -----------------------------
+    def list_dgljob_manifests(self, namespace: "str | None" = None):
+        """Raw DGLJob CR dicts (all namespaces unless one is given)."""
+        args = ["get", "dgljobs"]
+        args += ["-n", namespace] if namespace else ["-A"]
+        d = self._get_json(args)
+        if d is None:
+            return []
+        return d.get("items", [])
+
+    def update_dgljob_status(self, namespace: str, name: str,
+                             status: dict) -> bool:
+        """Write .status back (Status().Update parity,
+        dgljob_controller.go:304-315). Tries the status subresource first
+        (the CRD declares one); falls back to a plain merge patch."""
+        patch = json.dumps({"status": status})
+        r = self._run(["patch", "dgljob", name, "-n", namespace,
+                       "--subresource=status", "--type=merge", "-p", patch])
+        if r.returncode != 0:
+            r = self._run(["patch", "dgljob", name, "-n", namespace,
+                           "--type=merge", "-p", patch])
+        return r.returncode == 0

@@ -67,6 +67,9 @@ class Manager:
         self.interval = reconcile_interval
         self.ready = False
         self._stop = threading.Event()
+        # real-cluster mode: poll DGLJob CRs + write status back
+        self.watch_crs = hasattr(self.cluster, "list_dgljob_manifests")
+        self._written_statuses: Dict[str, dict] = {}
 
     # -- job API (the CRD surface) -----------------------------------------
     def submit(self, manifest) -> DGLJob:
@@ -82,8 +85,51 @@ class Manager:
     def get(self, namespace: str, name: str) -> Optional[DGLJob]:
         return self.jobs.get(f"{namespace}/{name}")
 
+    # -- DGLJob CR watch (real-cluster mode) -------------------------------
+    # The reference registers an informer on DGLJobs + owned Pods
+    # (main.go:73-105, dgljob_controller.go:436-458); here the same contract
+    # — every CR change observed, status written back — is met by polling
+    # the apiserver through the Cluster each loop iteration.
+    def sync_from_cluster(self):
+        lister = getattr(self.cluster, "list_dgljob_manifests", None)
+        if lister is None:
+            return
+        seen = set()
+        for item in lister():
+            meta = item.get("metadata", {})
+            key = f"{meta.get('namespace', 'default')}/{meta.get('name')}"
+            seen.add(key)
+            cur = self.jobs.get(key)
+            if cur is None:
+                self.jobs[key] = job_from_manifest(item)
+            else:
+                fresh = job_from_manifest(item)
+                # spec is mutable server-side; status/start-time are ours
+                cur.spec = fresh.spec
+                cur.uid = fresh.uid or cur.uid
+                if fresh.deletion_timestamp and not cur.deletion_timestamp:
+                    cur.deletion_timestamp = fresh.deletion_timestamp
+        # CRs deleted server-side: clean their pods then drop them
+        for key, job in list(self.jobs.items()):
+            if key not in seen and job.deletion_timestamp is None:
+                job.deletion_timestamp = time.time()
+
+    def _write_status(self, job: DGLJob):
+        writer = getattr(self.cluster, "update_dgljob_status", None)
+        if writer is None:
+            return
+        from .api import status_to_manifest
+
+        st = status_to_manifest(job)
+        prev = self._written_statuses.get(f"{job.namespace}/{job.name}")
+        if st != prev:  # only on change, like Status().Update on diff
+            if writer(job.namespace, job.name, st):
+                self._written_statuses[f"{job.namespace}/{job.name}"] = st
+
     # -- reconcile loop ----------------------------------------------------
     def reconcile_once(self):
+        if self.watch_crs:
+            self.sync_from_cluster()
         for key, job in list(self.jobs.items()):
             t0 = time.time()
             try:
@@ -107,6 +153,9 @@ class Manager:
                 continue
             if job.deletion_timestamp is not None:
                 del self.jobs[key]
+                self._written_statuses.pop(key, None)
+            elif self.watch_crs:
+                self._write_status(job)
 
     def run(self, metrics_port: Optional[int] = None,
             health_port: Optional[int] = None, block: bool = True):
@@ -144,11 +193,35 @@ def main(argv=None):
     p.add_argument("--watcher-loop-image", default="watcher-loop")
     p.add_argument("--kubectl-download-image", default="kubectl-download")
     p.add_argument("--job", action="append", default=[],
-                   help="DGLJob manifest YAML file(s) to manage")
+                   help="DGLJob manifest YAML file(s) to manage "
+                        "(local mode, no apiserver)")
+    p.add_argument("--kubectl", default=None, metavar="PATH",
+                   help="run against a real cluster: watch DGLJob CRs and "
+                        "reconcile through this kubectl binary")
     args = p.parse_args(argv)
-    mgr = Manager(reconcile_interval=args.reconcile_interval,
+    cluster = None
+    if args.kubectl or (not args.job):
+        # deployed mode (deploy/v1alpha1/dgl-operator.yaml runs with no
+        # --job files): drive the real apiserver via kubectl; fall back to
+        # the in-memory cluster only when kubectl cannot reach a cluster
+        from shutil import which
+
+        from .cluster import KubectlCluster
+
+        kubectl = args.kubectl or which("kubectl")
+        if kubectl:
+            probe = KubectlCluster(kubectl)
+            if probe._run(["version", "--request-timeout=5s"]).returncode == 0:
+                cluster = probe
+            elif args.kubectl:
+                raise SystemExit(
+                    f"--kubectl {args.kubectl}: cannot reach a cluster")
+    mgr = Manager(cluster=cluster,
+                  reconcile_interval=args.reconcile_interval,
                   watcher_loop_image=args.watcher_loop_image,
                   kubectl_download_image=args.kubectl_download_image)
+    mode = "cluster (DGLJob CR watch)" if mgr.watch_crs else "local --job files"
+    print(f"[manager] mode: {mode}", flush=True)
     for path in args.job:
         with open(path) as f:
             mgr.submit(f.read())

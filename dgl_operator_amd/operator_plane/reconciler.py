@@ -153,6 +153,7 @@ class DGLJobReconciler:
                 namespace=job.namespace,
                 data={"kubexec.sh": KUBEXEC_SH},
                 owner=job.name,
+                owner_uid=job.uid or None,
             )
             self.cluster.create_configmap(cm)
 
@@ -230,6 +231,7 @@ class DGLJobReconciler:
                            job.namespace, owner=job.name),
             ]
         for o in objs:
+            o.owner_uid = job.uid or None
             if c.get_rbac(o.namespace, o.kind, o.name) is None:
                 c.create_rbac(o)
 
@@ -294,6 +296,7 @@ class DGLJobReconciler:
                 "volumes": ["config", "kube", "dataset"],
             },
             owner=job.name,
+            owner_uid=job.uid or None,
         )
         self.cluster.create_pod(pod)
 
@@ -340,6 +343,7 @@ class DGLJobReconciler:
                 "shmSizeFraction": 0.5,
             },
             owner=job.name,
+            owner_uid=job.uid or None,
         )
 
     def _ensure_partitioner(self, job: DGLJob):
@@ -364,6 +368,7 @@ class DGLJobReconciler:
                     cluster_ip=None,  # headless
                     ports=list(range(DGL_PORT, DGL_PORT + HOST_PORT_NUM)),
                     owner=job.name,
+                    owner_uid=job.uid or None,
                 ))
 
     # -- status -----------------------------------------------------------

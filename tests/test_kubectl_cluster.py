@@ -1,0 +1,275 @@
+"""Integration tests for the real-cluster operator path: KubectlCluster +
+Manager DGLJob CR watch, driven through a file-backed fake kubectl.
+
+This is the path the reference exercises only by actually deploying
+(/root/reference/README.md:30-46; controller registered on the apiserver in
+main.go:73-105 and dgljob_controller.go:436-458). Here the whole chain —
+CR created with kubectl -> manager watch -> reconcile -> real v1 manifests
+created via kubectl -> kubelet transitions -> status written back to the CR
+-> CR deletion -> pod teardown — runs against tests/fake_kubectl.py.
+"""
+import json
+import os
+import stat
+import sys
+
+import pytest
+
+from dgl_operator_amd.operator_plane.api import (
+    JobPhase,
+    job_from_manifest,
+)
+from dgl_operator_amd.operator_plane.cluster import (
+    ConfigMap,
+    KubectlCluster,
+    Pod,
+    RBACObject,
+    Service,
+)
+from dgl_operator_amd.operator_plane.manager import Manager
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+
+
+@pytest.fixture()
+def kube(tmp_path, monkeypatch):
+    """(KubectlCluster, state_dir) against the fake kubectl."""
+    state = tmp_path / "kstate"
+    state.mkdir()
+    monkeypatch.setenv("FAKE_KUBE_STATE", str(state))
+    shim = tmp_path / "kubectl"
+    shim.write_text(
+        f"#!/bin/sh\nexec {sys.executable} {HERE}/fake_kubectl.py \"$@\"\n")
+    shim.chmod(shim.stat().st_mode | stat.S_IXUSR)
+    return KubectlCluster(kubectl=str(shim)), state
+
+
+def read_obj(state, kind, ns, name):
+    p = state / f"{kind}__{ns}__{name}.json"
+    if not p.exists():
+        return None
+    return json.loads(p.read_text())
+
+
+def write_obj(state, kind, ns, name, obj):
+    (state / f"{kind}__{ns}__{name}.json").write_text(json.dumps(obj))
+
+
+def set_pod_phase(state, ns, name, phase, ip=None, ready=None,
+                  reason=None):
+    """The fake kubelet: edits the stored pod status."""
+    obj = read_obj(state, "pod", ns, name)
+    assert obj is not None, f"no such pod {ns}/{name}"
+    st = obj.setdefault("status", {})
+    st["phase"] = phase
+    if ip:
+        st["podIP"] = ip
+    if reason:
+        st["reason"] = reason
+    if ready is None:
+        ready = phase == "Running"
+    for cs in st.setdefault("containerStatuses", [{"name": "c0"}]):
+        cs["ready"] = ready
+    write_obj(state, "pod", ns, name, obj)
+
+
+def test_object_round_trips(kube):
+    c, state = kube
+    cm = ConfigMap("j-config", "ns1", data={"kubexec.sh": "#!/bin/sh\n",
+                                            "hostfile": ""})
+    c.create_configmap(cm)
+    got = c.get_configmap("ns1", "j-config")
+    assert got.data["kubexec.sh"].startswith("#!")
+    cm.data["hostfile"] = "1.2.3.4 30050 w-0 slots=1\n"
+    c.update_configmap(cm)
+    assert "1.2.3.4" in c.get_configmap("ns1", "j-config").data["hostfile"]
+
+    pod = Pod("j-worker-0", "ns1", labels={"dgl-job-name": "j"},
+              spec={"containers": [{"name": "w", "image": "img"}],
+                    "volumes": ["config", "shm"]},
+              owner="j")
+    c.create_pod(pod)
+    got = c.get_pod("ns1", "j-worker-0")
+    assert got is not None and got.phase.value == "Pending"
+    assert not got.is_real_running()
+    # duplicate create is AlreadyExists -> swallowed (reconciler gets first)
+    c.create_pod(pod)
+    assert [p.name for p in c.list_pods("ns1", "j")] == ["j-worker-0"]
+
+    c.create_service(Service("j-worker-0", "ns1",
+                             selector={"dgl-replica-name": "j-worker-0"},
+                             cluster_ip=None, ports=[30050, 30051]))
+    svc = c.get_service("ns1", "j-worker-0")
+    assert svc.cluster_ip is None and svc.ports == [30050, 30051]
+
+    c.create_rbac(RBACObject("ServiceAccount", "j-launcher", "ns1"))
+    c.create_rbac(RBACObject(
+        "Role", "j-launcher", "ns1",
+        rules=[{"resources": ["pods"], "verbs": ["get", "list", "watch"]},
+               {"resources": ["pods/exec"], "verbs": ["create"],
+                "resourceNames": ["j-worker-0"]}]))
+    c.create_rbac(RBACObject("RoleBinding", "j-launcher", "ns1"))
+    role = c.get_rbac("ns1", "Role", "j-launcher")
+    assert role.rules[1]["resourceNames"] == ["j-worker-0"]
+    assert c.get_rbac("ns1", "ServiceAccount", "j-launcher") is not None
+
+    c.delete_pod("ns1", "j-worker-0")
+    assert c.get_pod("ns1", "j-worker-0") is None
+
+
+def _submit_cr(c, state, path):
+    with open(path) as f:
+        manifest_text = f.read()
+    import yaml
+
+    manifest = yaml.safe_load(manifest_text)
+    r = c._run(["create", "-f", "-"], stdin=json.dumps(manifest))
+    assert r.returncode == 0, r.stderr
+    return manifest
+
+
+def test_full_ladder_through_kubectl(kube):
+    """DGLJob CR -> manager watch -> real manifests -> phase ladder ->
+    status writeback -> CR deletion -> teardown."""
+    c, state = kube
+    manifest = _submit_cr(
+        c, state,
+        os.path.join(HERE, "..", "examples", "v1alpha1",
+                     "GraphSAGE_dist.yaml"))
+    ns = manifest["metadata"]["namespace"]
+    name = manifest["metadata"]["name"]
+    n_workers = manifest["spec"]["dglReplicaSpecs"]["Worker"]["replicas"]
+
+    mgr = Manager(cluster=c)
+    assert mgr.watch_crs
+    mgr.reconcile_once()
+    job = mgr.get(ns, name)
+    assert job is not None and job.uid  # uid came from the apiserver
+
+    # launcher + partitioner exist as REAL manifests
+    lp = read_obj(state, "pod", ns, f"{name}-launcher")
+    pp = read_obj(state, "pod", ns, f"{name}-partitioner")
+    assert lp and pp
+    assert lp["spec"]["restartPolicy"] == "Never"
+    inits = [ic["name"] for ic in lp["spec"]["initContainers"]]
+    assert inits == ["kubectl-download", "watcher-loop-partitioner",
+                     "watcher-loop-worker"]
+    # watcher env carries mounted paths
+    wl = lp["spec"]["initContainers"][1]
+    env = {e["name"]: e["value"] for e in wl["env"]}
+    assert env["WATCHERFILE"] == "/etc/dgl/partfile"
+    assert env["WATCHERMODE"] == "finished"
+    assert env["NAMESPACE"] == ns
+    # partitioner-watcher mounts the dataset emptyDir (copy-target trick)
+    assert any(m["mountPath"] == "/dgl_workspace/dataset"
+               for m in wl["volumeMounts"])
+    # ownerReferences attached for apiserver GC
+    assert lp["metadata"]["ownerReferences"][0]["kind"] == "DGLJob"
+    assert lp["metadata"]["ownerReferences"][0]["uid"] == job.uid
+    # config volume with mode-bearing items
+    vols = {v["name"]: v for v in lp["spec"]["volumes"]}
+    items = {i["key"]: i["mode"] for i in vols["config-volume"]["configMap"]["items"]}
+    assert items["kubexec.sh"] == 0o555 and items["hostfile"] == 0o444
+    # status written back to the CR (partitioner already created but not
+    # yet running -> Starting)
+    cr = read_obj(state, "dgljob", ns, name)
+    assert cr["status"]["phase"] == JobPhase.STARTING.value
+
+    # kubelet: partitioner runs -> Partitioning
+    set_pod_phase(state, ns, f"{name}-partitioner", "Running", ip="10.0.0.2")
+    set_pod_phase(state, ns, f"{name}-launcher", "Pending", ready=False)
+    mgr.reconcile_once()
+    assert read_obj(state, "dgljob", ns, name)["status"]["phase"] == \
+        JobPhase.PARTITIONING.value
+    # partfile got the partitioner IP
+    cm = read_obj(state, "configmap", ns, f"{name}-config")
+    assert cm["data"]["partfile"].startswith("10.0.0.2 30050")
+
+    # partitioner done -> Partitioned; workers + headless services created
+    set_pod_phase(state, ns, f"{name}-partitioner", "Succeeded", ready=False)
+    mgr.reconcile_once()
+    mgr.reconcile_once()
+    for i in range(n_workers):
+        wp = read_obj(state, "pod", ns, f"{name}-worker-{i}")
+        assert wp, f"worker {i} missing"
+        # sleep-365d default command, 20 ports, shm emptyDir
+        assert wp["spec"]["containers"][0]["command"] == ["sleep", "365d"] or \
+            "command" in wp["spec"]["containers"][0]
+        ports = wp["spec"]["containers"][0]["ports"]
+        assert len(ports) == 20 and ports[0]["containerPort"] == 30050
+        wvols = {v["name"]: v for v in wp["spec"]["volumes"]}
+        assert wvols["dshm"]["emptyDir"]["medium"] == "Memory"
+        svc = read_obj(state, "service", ns, f"{name}-worker-{i}")
+        assert svc["spec"]["clusterIP"] == "None"
+        assert svc["spec"]["selector"]["dgl-replica-name"] == \
+            f"{name}-worker-{i}"
+    assert read_obj(state, "dgljob", ns, name)["status"]["phase"] == \
+        JobPhase.PARTITIONED.value
+
+    # workers + launcher run -> Training; hostfile filled with worker IPs
+    for i in range(n_workers):
+        set_pod_phase(state, ns, f"{name}-worker-{i}", "Running",
+                      ip=f"10.0.1.{i}")
+    set_pod_phase(state, ns, f"{name}-launcher", "Running", ip="10.0.0.9")
+    mgr.reconcile_once()
+    cr = read_obj(state, "dgljob", ns, name)
+    assert cr["status"]["phase"] == JobPhase.TRAINING.value
+    assert cr["status"]["replicaStatuses"]["Worker"]["ready"] == \
+        f"{n_workers}/{n_workers}"
+    cm = read_obj(state, "configmap", ns, f"{name}-config")
+    lines = cm["data"]["hostfile"].strip().splitlines()
+    assert len(lines) == n_workers
+    assert lines[0].split() == ["10.0.1.0", "30050", f"{name}-worker-0",
+                                "slots=1"]
+
+    # launcher succeeds -> Completed (+ completionTime stamped)
+    set_pod_phase(state, ns, f"{name}-launcher", "Succeeded", ready=False)
+    mgr.reconcile_once()
+    cr = read_obj(state, "dgljob", ns, name)
+    assert cr["status"]["phase"] == JobPhase.COMPLETED.value
+    assert "completionTime" in cr["status"]
+
+    # CR deleted server-side -> cleanup deletes the pods, job dropped
+    os.unlink(state / f"dgljob__{ns}__{name}.json")
+    mgr.reconcile_once()
+    mgr.reconcile_once()
+    assert read_obj(state, "pod", ns, f"{name}-worker-0") is None
+    assert mgr.get(ns, name) is None
+
+
+def test_skip_mode_and_failure_through_kubectl(kube):
+    c, state = kube
+    manifest = _submit_cr(
+        c, state,
+        os.path.join(HERE, "..", "examples", "v1alpha1", "GraphSAGE.yaml"))
+    ns, name = (manifest["metadata"]["namespace"],
+                manifest["metadata"]["name"])
+    mgr = Manager(cluster=c)
+    mgr.reconcile_once()
+    lp = read_obj(state, "pod", ns, f"{name}-launcher")
+    # Skip mode: Launcher_Workload phase env, no partitioner watcher
+    env = {e["name"]: e["value"]
+           for e in lp["spec"]["containers"][0]["env"]}
+    assert env["DGL_OPERATOR_PHASE_ENV"] == "Launcher_Workload"
+    inits = [ic["name"] for ic in lp["spec"].get("initContainers", [])]
+    assert "watcher-loop-partitioner" not in inits
+    assert read_obj(state, "pod", ns, f"{name}-partitioner") is None
+
+    # a failed pod fails the job and the status lands on the CR
+    set_pod_phase(state, ns, f"{name}-launcher", "Failed", ready=False)
+    mgr.reconcile_once()
+    assert read_obj(state, "dgljob", ns, name)["status"]["phase"] == \
+        JobPhase.FAILED.value
+
+
+def test_job_from_manifest_uid_and_deletion():
+    m = {
+        "apiVersion": "qihoo.net/v1alpha1", "kind": "DGLJob",
+        "metadata": {"name": "x", "namespace": "d", "uid": "u-1",
+                     "deletionTimestamp": "2026-01-01T00:00:00Z"},
+        "spec": {"partitionMode": "Skip",
+                 "dglReplicaSpecs": {"Launcher": {"replicas": 1}}},
+    }
+    job = job_from_manifest(m)
+    assert job.uid == "u-1"
+    assert job.deletion_timestamp is not None
