@@ -61,6 +61,28 @@ def build_parser():
     p.add_argument("--num-servers", type=int, default=0,
                    help="accepted for reference-CLI parity; this framework "
                         "has no server processes")
+    # flags consumed by examples/v1alpha1/GraphSAGE_dist.yaml verbatim
+    # (reference dglrun:2-105): trainers become torchrun ranks per pod;
+    # samplers/servers are collapsed into the trainer process (the GPU
+    # sampler is a HIP kernel, the feature store is HBM + RCCL alltoallv)
+    p.add_argument("--num-trainers", type=int, default=None,
+                   help="trainer ranks per worker pod (default: hostfile "
+                        "slots)")
+    p.add_argument("--num-samplers", type=int, default=0,
+                   help="accepted for reference-CLI parity; sampling runs "
+                        "in-process on the GPU (no sampler processes)")
+    p.add_argument("--num-workers", type=int, default=None,
+                   help="expected worker-pod count; validated against the "
+                        "hostfile when set")
+    p.add_argument("--dataset-url", default="",
+                   help="forwarded to the partition entry point "
+                        "(--dataset-url); offline partitioners fall back "
+                        "to synthetic data")
+    p.add_argument("--launch-entry-point", default="",
+                   help="accepted for reference-CLI parity (built-in "
+                        "launcher is used)")
+    p.add_argument("--revise-hostfile-entry-point", default="",
+                   help="custom hostfile-revise script (default: built-in)")
     p.add_argument("--workspace", "--worksapce", default=os.environ.get(
         "WORKSPACE", "/dgl_workspace"))
     p.add_argument("--hostfile", default="/etc/dgl/hostfile")
@@ -91,6 +113,8 @@ def run_phase1(args, dataset):
             extra += " --no-balance-train"
         if not args.balance_edges:
             extra += " --no-balance-edges"
+        if args.dataset_url:
+            extra += f" --dataset-url {args.dataset_url}"
         _run(
             f"python {args.partition_entry_point} "
             f"--graph-name {args.graph_name} "
@@ -127,6 +151,10 @@ def _deliver(args, dataset):
 def run_launcher(args):
     with open(args.hostfile) as f:
         hosts = parse_hostfile(f.read())
+    if args.num_workers is not None and args.num_workers != len(hosts):
+        raise SystemExit(
+            f"[dglrun] --num-workers {args.num_workers} but hostfile has "
+            f"{len(hosts)} workers")
     dataset = os.path.join(args.workspace, "dataset")
     with phase("Phase 3/5 dispatch"):
         if args.dispatch_entry_point:
@@ -137,12 +165,19 @@ def run_launcher(args):
             dispatch_partitions(dataset, args.graph_name, hosts,
                                 workspace=args.workspace)
     with phase("Phase 4/5 revise hostfile"):
-        revised = revise_for_dgl(hosts)
-        launch_mod.exec_batch(
-            hosts,
-            f"sh -c 'mkdir -p {args.workspace} && printf %s \"{revised}\" "
-            f"> {args.workspace}/hostfile_revised'",
-        )
+        if args.revise_hostfile_entry_point:
+            launch_mod.exec_batch(
+                hosts,
+                f"python {args.revise_hostfile_entry_point} "
+                f"--hostfile {args.hostfile} --workspace {args.workspace}",
+            )
+        else:
+            revised = revise_for_dgl(hosts)
+            launch_mod.exec_batch(
+                hosts,
+                f"sh -c 'mkdir -p {args.workspace} && printf %s \"{revised}\" "
+                f"> {args.workspace}/hostfile_revised'",
+            )
     with phase("Phase 5/5 train"):
         part_cfg = (args.partition_config_path
                     or f"{args.workspace}/workload/{args.graph_name}.json")
@@ -157,6 +192,7 @@ def run_launcher(args):
             f"--graph-name {args.graph_name} "
             f"--part-config {part_cfg} {targs}",
             master_port=args.master_port,
+            num_trainers=args.num_trainers,
         )
 
 

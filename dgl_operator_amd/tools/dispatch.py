@@ -2,9 +2,14 @@
 its output; rewrites per-part paths into the workers' workload dir and ships
 each part to its worker over the fabric.
 
-Reference: /root/reference/python/dglrun/tools/dispatch.py:52-91 (one
-partition per worker is enforced there at launch.py:107-108 — same check
-here).
+Reference: /root/reference/python/dglrun/tools/dispatch.py:52-91. The
+reference enforces one partition per worker POD (launch.py:107-108) because
+its slots only multiply trainer processes over a shared partition; here
+slots = ranks = GPUs per pod and the invariant generalizes to ONE PARTITION
+PER RANK: pod i receives the contiguous partition block
+[sum(slots[:i]), sum(slots[:i+1])), and torchrun global rank r trains
+part-r. With slots=1 everywhere this reduces exactly to the reference
+behavior.
 """
 from __future__ import annotations
 
@@ -31,9 +36,10 @@ def dispatch_partitions(
     with open(meta_path) as f:
         meta = json.load(f)
     num_parts = meta["num_parts"]
-    assert num_parts == len(hosts), (
-        f"one partition per worker required: {num_parts} parts vs "
-        f"{len(hosts)} workers"
+    total_ranks = sum(h.slots for h in hosts)
+    assert num_parts == total_ranks, (
+        f"one partition per rank required: {num_parts} parts vs "
+        f"{total_ranks} ranks ({len(hosts)} pods x slots)"
     )
     revised = dict(meta)
     for i in range(num_parts):
@@ -46,15 +52,18 @@ def dispatch_partitions(
     with open(revised_path, "w") as f:
         json.dump(revised, f, indent=2)
 
-    for i, host in enumerate(hosts):
-        part_src = os.path.join(dataset_dir, f"part{i}")
+    part = 0
+    for host in hosts:
         dst_dir = f"{workspace}/{WORKLOAD_DIR}"
-        fabric.exec(host.pod, f"mkdir -p {dst_dir}/part{i}")
-        for fname in os.listdir(part_src):
-            fabric.copy(
-                os.path.join(part_src, fname), host.pod,
-                f"{dst_dir}/part{i}/{fname}",
-            )
+        for _ in range(host.slots):
+            part_src = os.path.join(dataset_dir, f"part{part}")
+            fabric.exec(host.pod, f"mkdir -p {dst_dir}/part{part}")
+            for fname in os.listdir(part_src):
+                fabric.copy(
+                    os.path.join(part_src, fname), host.pod,
+                    f"{dst_dir}/part{part}/{fname}",
+                )
+            part += 1
         fabric.copy(revised_path, host.pod, f"{dst_dir}/{graph_name}.json")
     return revised
 

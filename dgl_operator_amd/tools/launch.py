@@ -51,18 +51,21 @@ def train(
     master_port: int = 29400,
     fabric: Optional[Fabric] = None,
     env: Optional[Dict[str, str]] = None,
+    num_trainers: Optional[int] = None,
 ):
-    """Launch one torchrun per worker pod; rank-per-GPU = slots; rendezvous at
-    worker 0's pod IP (the operator's hostfile carries the IPs)."""
+    """Launch one torchrun per worker pod; ranks per pod = slots (or the
+    dglrun --num-trainers override, reference launch.py:135-152); rendezvous
+    at worker 0's pod IP (the operator's hostfile carries the IPs)."""
     fabric = fabric or get_fabric()
     assert hosts, "no workers in hostfile"
     master = hosts[0].ip
     nnodes = len(hosts)
     procs = []
     for i, h in enumerate(hosts):
+        nproc = num_trainers if num_trainers else h.slots
         cmd = (
             f"python -m torch.distributed.run --nnodes={nnodes} "
-            f"--node-rank={i} --nproc-per-node={h.slots} "
+            f"--node-rank={i} --nproc-per-node={nproc} "
             f"--master-addr={master} --master-port={master_port} "
             f"{script} {script_args}"
         )

@@ -24,6 +24,10 @@ def main():
     p.add_argument("--graph-name", default="ogbn-products")
     p.add_argument("--num-partitions", type=int, required=True)
     p.add_argument("--output", required=True)
+    p.add_argument("--dataset-url", default="",
+                   help="dataset archive URL (reference manifest parity, "
+                        "GraphSAGE_dist.yaml:30-31); offline environments "
+                        "log it and fall back to --dataset synthesis")
     p.add_argument("--dataset", default="rmat",
                    help="rmat | path to a .pt with {src,dst,feat,label}")
     p.add_argument("--nodes", type=int, default=2_449_029)
