@@ -71,6 +71,25 @@ def test_dispatch_over_local_fabric(tmp_path):
         assert meta[f"part-{i}"]["part_graph"] == f"part{i}/graph.pt"
 
 
+def test_dispatch_slots_gt_one_partitions_per_rank(tmp_path):
+    """slotsPerWorker=2: pod i receives the contiguous partition block for
+    its ranks (one partition per rank, 2 pods x 2 slots = 4 parts)."""
+    g = rmat_graph(80, 600, num_feats=4, seed=4)
+    ds = tmp_path / "dataset"
+    partition_graph(g, "toy", 4, str(ds), algorithm="range")
+    fab = LocalFabric(str(tmp_path / "pods"))
+    hosts = parse_hostfile(
+        "10.244.0.5 30050 job-worker-0 slots=2\n"
+        "10.244.0.6 30050 job-worker-1 slots=2\n"
+    )
+    dispatch_partitions(str(ds), "toy", hosts, fabric=fab, workspace="/w")
+    for pod, parts in (("job-worker-0", (0, 1)), ("job-worker-1", (2, 3))):
+        pd = fab.pod_dir(pod)
+        for p in parts:
+            assert os.path.exists(f"{pd}/w/workload/part{p}/graph.pt"), (pod, p)
+        assert os.path.exists(f"{pd}/w/workload/toy.json")
+
+
 def test_dispatch_rejects_mismatched_counts(tmp_path):
     g = rmat_graph(60, 400, seed=3)
     ds = tmp_path / "dataset"
@@ -81,7 +100,7 @@ def test_dispatch_rejects_mismatched_counts(tmp_path):
         dispatch_partitions(str(ds), "toy", hosts, fabric=fab)
         assert False, "should have raised"
     except AssertionError as e:
-        assert "one partition per worker" in str(e)
+        assert "one partition per rank" in str(e)
 
 
 def test_dist_graph_from_partition(tmp_path):
