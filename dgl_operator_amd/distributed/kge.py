@@ -33,6 +33,8 @@ class DistKGEModel:
         world_size: int = 1,
         device="cpu",
         seed: int = 0,
+        entity_boundaries=None,
+        relation_boundaries=None,
     ):
         self.num_entities = num_entities
         self.hidden_dim = hidden_dim
@@ -40,7 +42,7 @@ class DistKGEModel:
         self.score = get_score_func(score_func, gamma=gamma, emb_init=emb_init)
         self.entities = ShardedEmbedding(
             num_entities, hidden_dim, world_size, rank, device=device,
-            init_range=emb_init, seed=seed,
+            init_range=emb_init, seed=seed, boundaries=entity_boundaries,
         )
         rel_dim = hidden_dim
         if score_func == "RotatE":
@@ -54,6 +56,7 @@ class DistKGEModel:
         self.relations = ShardedEmbedding(
             num_relations, rel_dim, world_size, rank, device=device,
             init_range=emb_init, seed=seed + 7,
+            boundaries=relation_boundaries,
         )
 
     def train_step(

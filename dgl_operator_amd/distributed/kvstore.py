@@ -31,8 +31,18 @@ class ShardedEmbedding:
         device="cpu",
         init_range: float = 1.0,
         seed: int = 0,
+        boundaries=None,
     ):
-        bounds = [num_rows * p // num_parts for p in range(num_parts + 1)]
+        # boundaries: explicit (possibly uneven) range bounds, e.g. the
+        # degree-balanced entity partition written by tools/kg_partition.py;
+        # default = even split
+        if boundaries is not None:
+            bounds = [int(b) for b in boundaries]
+            assert len(bounds) == num_parts + 1 and bounds[-1] == num_rows, (
+                f"bad boundaries {bounds} for {num_rows} rows / "
+                f"{num_parts} parts")
+        else:
+            bounds = [num_rows * p // num_parts for p in range(num_parts + 1)]
         self.book = PartitionBook(bounds, device=device)
         self.rank = rank
         self.lo, self.hi = self.book.owned_range(rank)

@@ -62,16 +62,103 @@ def build_parser():
     p.add_argument("--workspace", "--worksapce", default=os.environ.get(
         "WORKSPACE", "/dgl_workspace"))
     p.add_argument("--hostfile", default="/etc/dgl/hostfile")
+    p.add_argument("--leadfile", default="/etc/dgl/leadfile")
     p.add_argument("--master-port", type=int, default=29401)
+    p.add_argument("--num-triples", type=int, default=100_000)
+    p.add_argument("--num-entities", type=int, default=10_000)
+    p.add_argument("--num-relations", type=int, default=100)
     p.add_argument("--train-entry-point",
                    default="examples/dgl_ke/train_ke.py")
     return p
 
 
+def _dataset_name(args) -> str:
+    return args.custom_dataset or args.dataset
+
+
+def run_partitioner(args):
+    """Phases 1-2 on the partitioner pod (reference dglkerun:133-205):
+    partition the KG (built-in synthetic or custom triple files), then
+    deliver the partitioned dataset into the launcher's still-waiting
+    watcher-loop-partitioner init container."""
+    from .fabric import get_fabric
+    from . import kg_partition
+
+    dataset_root = os.path.join(args.workspace, "dataset")
+    os.makedirs(dataset_root, exist_ok=True)
+    name = _dataset_name(args)
+    meta_json = os.path.join(dataset_root, name, f"{name}.json")
+    if args.ignore_partition and os.path.exists(meta_json):
+        print("[dglkerun] Phase 1/5 skipped (--ignore-partition)", flush=True)
+    else:
+        with phase("Phase 1/5 partition KG"):
+            argv = ["--dataset", name, "-k", str(args.num_partitions or 1),
+                    "--data-path", dataset_root]
+            if args.dataset_files:
+                argv += ["--data-files"] + args.dataset_files.split(",")
+                if args.dataset_format:
+                    argv += ["--format", args.dataset_format]
+            else:
+                argv += ["--num-entities", str(args.num_entities),
+                         "--num-relations", str(args.num_relations),
+                         "--num-triples", str(args.num_triples)]
+            kg_partition.main(argv)
+    with phase("Phase 2/5 deliver"):
+        if args.partitioned_dataset_dir or args.pvc_partitioned_dir:
+            # PVC path (reference dglkerun:190-205 scp branch): the shared
+            # volume already holds the partition; nothing to copy
+            print("[dglkerun] PVC-partitioned dir in use; skip delivery",
+                  flush=True)
+            return
+        leads = []
+        deadline = time.time() + 120
+        while time.time() < deadline:
+            try:
+                with open(args.leadfile) as f:
+                    leads = parse_hostfile(f.read())
+            except FileNotFoundError:
+                leads = []
+            if leads:
+                break
+            time.sleep(0.5)
+        if not leads:
+            raise SystemExit("[dglkerun] no launcher entry in leadfile")
+        fabric = get_fabric()
+        for lead in leads:
+            fabric.copy(os.path.join(dataset_root, name), lead.pod,
+                        f"{args.workspace}/dataset/{name}",
+                        container="watcher-loop-partitioner")
+
+
 def main(argv=None):
     args = build_parser().parse_args(argv)
+    mode = os.environ.get("DGL_OPERATOR_PHASE_ENV", "")
+    if mode == "Launcher_Workload":
+        with phase("workload (Skip mode)"):
+            import subprocess as sp
+
+            rc = sp.call(f"python {args.train_entry_point}", shell=True)
+            if rc != 0:
+                raise SystemExit(rc)
+        return
+    if mode == "Partitioner":
+        run_partitioner(args)
+        return
     with open(args.hostfile) as f:
         hosts = parse_hostfile(f.read())
+    name = _dataset_name(args)
+    dataset_dir = (args.partitioned_dataset_dir or args.pvc_partitioned_dir
+                   or os.path.join(args.workspace, "dataset"))
+    have_partition = os.path.exists(
+        os.path.join(dataset_dir, name, f"{name}.json"))
+    if have_partition:
+        # Phase 3 (reference dglkerun:225-233): the WHOLE partitioned
+        # dataset goes to every worker (each rank reads its own part file)
+        with phase("Phase 3/5 dispatch (copy dataset to workers)"):
+            launch_mod.copy_batch(
+                hosts, os.path.join(dataset_dir, name),
+                f"{args.workspace}/dataset/{name}",
+            )
     with phase("Phase 4/5 revise hostfile (dglke format)"):
         revised = revise_for_dglke(hosts, num_servers=1)
         launch_mod.exec_batch(
@@ -88,6 +175,13 @@ def main(argv=None):
             f"--max-step {args.max_step} --save-path {args.save_path}"
             + (" --no-save-emb" if args.no_save_emb else "")
         )
+        if have_partition:
+            # PVC mode: workers read the shared dir; copy mode: the Phase 3
+            # copy landed under the per-pod workspace
+            dp = (dataset_dir
+                  if (args.partitioned_dataset_dir or args.pvc_partitioned_dir)
+                  else f"{args.workspace}/dataset")
+            extra += f" --data-path {dp} --dataset-name {name}"
         launch_mod.train(hosts, args.train_entry_point, extra,
                          master_port=args.master_port)
 

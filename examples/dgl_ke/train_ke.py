@@ -110,6 +110,12 @@ def main():
     p.add_argument("--num-entities", type=int, default=1_000_000)
     p.add_argument("--num-relations", type=int, default=1000)
     p.add_argument("--num-triples", type=int, default=5_000_000)
+    p.add_argument("--data-path", default="",
+                   help="partitioned dataset root (tools/kg_partition.py "
+                        "output); rank r trains part-r's triples and the "
+                        "shard boundaries come from the partition json")
+    p.add_argument("--dataset-name", default="synthetic",
+                   help="dataset dir name under --data-path")
     p.add_argument("--save-path", default="")
     p.add_argument("--no-save-emb", action="store_true")
     p.add_argument("--no-capture", dest="capture", action="store_false",
@@ -132,19 +138,41 @@ def main():
     else:
         device = torch.device("cpu")
 
-    gen = torch.Generator(device=device)
-    gen.manual_seed(0)
-    h = torch.randint(0, args.num_entities, (args.num_triples,),
-                      generator=gen, device=device)
-    r = torch.randint(0, args.num_relations, (args.num_triples,),
-                      generator=gen, device=device)
-    t = torch.randint(0, args.num_entities, (args.num_triples,),
-                      generator=gen, device=device)
+    ent_bounds = rel_bounds = None
+    if args.data_path:
+        # partitioned dataset (dglkerun Phases 1-3 output): rank r owns
+        # part-r's triples; entity/relation shard boundaries from the json
+        root = os.path.join(args.data_path, args.dataset_name)
+        with open(os.path.join(root, f"{args.dataset_name}.json")) as f:
+            meta = json.load(f)
+        assert meta["num_parts"] == ws, (
+            f"partitioned for {meta['num_parts']} ranks but WORLD_SIZE={ws}")
+        args.num_entities = meta["num_entities"]
+        args.num_relations = meta["num_relations"]
+        ent_bounds = meta["entity_boundaries"]
+        rel_bounds = meta["relation_boundaries"]
+        triples = torch.load(
+            os.path.join(root, meta["parts"][str(rank)]["train"]),
+            weights_only=True).to(device)
+        h, r, t = triples[:, 0], triples[:, 1], triples[:, 2]
+        print(f"[train_ke] rank {rank}: {h.numel()} triples, "
+              f"entities {ent_bounds[rank]}..{ent_bounds[rank + 1]}",
+              flush=True)
+    else:
+        gen = torch.Generator(device=device)
+        gen.manual_seed(0)
+        h = torch.randint(0, args.num_entities, (args.num_triples,),
+                          generator=gen, device=device)
+        r = torch.randint(0, args.num_relations, (args.num_triples,),
+                          generator=gen, device=device)
+        t = torch.randint(0, args.num_entities, (args.num_triples,),
+                          generator=gen, device=device)
 
     model = DistKGEModel(
         args.num_entities, args.num_relations, args.hidden_dim,
         score_func=args.model_name, gamma=args.gamma, rank=rank,
         world_size=ws, device=device,
+        entity_boundaries=ent_bounds, relation_boundaries=rel_bounds,
     )
     sampler = KGEdgeSampler(
         (h, r, t), args.num_entities, batch_size=args.batch_size,
@@ -207,8 +235,31 @@ def main():
     if args.eval:
         from dgl_operator_amd.distributed.kge import evaluate_kge
 
-        m = evaluate_kge(model, h[: args.num_eval], r[: args.num_eval],
-                         t[: args.num_eval])
+        eh, er, et = h, r, t
+        if args.data_path:
+            # every rank must evaluate IDENTICAL triples (the eval pulls are
+            # collectives): use the global valid split when present,
+            # otherwise rank 0's local slice broadcast to everyone
+            root = os.path.join(args.data_path, args.dataset_name)
+            vpath = os.path.join(root, "valid.pt")
+            if os.path.exists(vpath):
+                v = torch.load(vpath, weights_only=True).to(device)
+                eh, er, et = v[:, 0], v[:, 1], v[:, 2]
+            elif ws > 1:
+                import torch.distributed as dist
+
+                n = torch.tensor([min(args.num_eval, eh.numel())],
+                                 device=device)
+                dist.broadcast(n, src=0)
+                buf = torch.zeros(3, int(n[0]), dtype=torch.int64,
+                                  device=device)
+                if rank == 0:
+                    buf.copy_(torch.stack([eh[: int(n[0])], er[: int(n[0])],
+                                           et[: int(n[0])]]))
+                dist.broadcast(buf, src=0)
+                eh, er, et = buf[0], buf[1], buf[2]
+        m = evaluate_kge(model, eh[: args.num_eval], er[: args.num_eval],
+                         et[: args.num_eval])
         if rank == 0:
             print("eval:", {k: round(v, 4) for k, v in m.items()}, flush=True)
     if args.save_path and not args.no_save_emb:
