@@ -199,6 +199,16 @@ def main():
     rank, ws = comm.init_from_env()
     if ws > 1 and args.gpus != ws and rank == 0:
         print(f"# note: --gpus {args.gpus} overridden by WORLD_SIZE={ws}")
+    if args.capture_dist:
+        # fail fast on unsupported combinations instead of silently
+        # benching a different configuration than asked
+        if args.no_halo:
+            raise SystemExit("--capture-dist requires the halo path "
+                             "(remove --no-halo): the captured step records "
+                             "only the gradient all-reduce collective")
+        if len(args.fanout.split(",")) != 2:
+            raise SystemExit("--capture-dist supports the 2-layer flagship "
+                             "config only (fanout must have 2 entries)")
     if args.device:
         device = torch.device(args.device)
     elif torch.cuda.is_available():
