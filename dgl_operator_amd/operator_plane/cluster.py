@@ -114,6 +114,7 @@ class FakeCluster(Cluster):
         self.services: Dict[str, Service] = {}
         self.rbac: Dict[str, RBACObject] = {}
         self._next_ip = 2
+        self._watchers: List[Any] = []  # event queues (watch_pods)
 
     @staticmethod
     def _key(namespace: str, name: str) -> str:
@@ -152,6 +153,27 @@ class FakeCluster(Cluster):
         if containers_ready is None:
             containers_ready = phase == PodPhase.RUNNING
         p.containers_ready = containers_ready
+        for q in self._watchers:  # pod UPDATE event to every subscriber
+            q.put(p)
+
+    def watch_pods(self, namespace: str):
+        """Pod event stream (informer UpdateFunc analog): yields a Pod on
+        every phase transition, None as a 0.2 s keep-alive heartbeat."""
+        import queue as _queue
+
+        q: "_queue.Queue" = _queue.Queue()
+        self._watchers.append(q)
+        try:
+            while True:
+                try:
+                    ev = q.get(timeout=0.2)
+                except _queue.Empty:
+                    yield None
+                    continue
+                if ev.namespace == namespace:
+                    yield ev
+        finally:
+            self._watchers.remove(q)
 
     def run_all_pending(self, namespace: Optional[str] = None):
         for p in list(self.pods.values()):
@@ -332,6 +354,39 @@ class KubectlCluster(Cluster):
 
         self._create(rbac_manifest(obj))
         return obj
+
+    def watch_pods(self, namespace: str):
+        """Pod event stream over ``kubectl get pods --watch -o name``
+        (the informer analog for real clusters): each change event names a
+        pod, whose state is then fetched — robust against arbitrary JSON
+        in annotations, at the cost of one extra get per event. Yields a
+        Pod per event, None heartbeats while the stream is quiet."""
+        import select
+
+        proc = subprocess.Popen(
+            [self.kubectl, "get", "pods", "-n", namespace, "--watch",
+             "-o", "name"],
+            stdout=subprocess.PIPE, stderr=subprocess.DEVNULL, text=True,
+        )
+        try:
+            while True:
+                r, _, _ = select.select([proc.stdout], [], [], 0.5)
+                if not r:
+                    if proc.poll() is not None:
+                        return
+                    yield None
+                    continue
+                line = proc.stdout.readline()
+                if not line:
+                    return
+                name = line.strip().rsplit("/", 1)[-1]
+                if not name:
+                    continue
+                pod = self.get_pod(namespace, name)
+                if pod is not None:
+                    yield pod
+        finally:
+            proc.terminate()
 
     # -- DGLJob CRs (the Manager watch surface) -----------------------------
     def list_dgljob_manifests(self, namespace: "str | None" = None):

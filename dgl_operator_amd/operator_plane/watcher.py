@@ -67,6 +67,60 @@ def watch_file(cluster: Cluster, namespace: str, watchfile_text: str,
     return watch(cluster, namespace, parse_watchfile(watchfile_text), mode, **kw)
 
 
+def _satisfied(pod, mode: str) -> bool:
+    if mode == "ready":
+        return pod.is_real_running()
+    return pod.phase == PodPhase.SUCCEEDED
+
+
+def watch_events(
+    cluster: Cluster,
+    namespace: str,
+    pod_names: Iterable[str],
+    mode: str = "ready",
+    timeout: Optional[float] = None,
+) -> bool:
+    """Event-driven variant of :func:`watch` — the informer + workqueue
+    shape of the reference watcher-loop
+    (/root/reference/watcher-loop/controllers/controller.go:84-152):
+    an initial list sweep (:121-132), then pod UPDATE events filtered on
+    the watched names (:84-100) drain the set; returns when it is empty
+    (:140-152). No steady-state polling of the apiserver.
+
+    Requires the cluster to expose ``watch_pods(namespace)`` — a generator
+    of Pod events (``None`` items are keep-alive heartbeats used for the
+    timeout check). Falls back to :func:`watch` when it does not.
+    """
+    assert mode in ("ready", "finished")
+    stream_fn = getattr(cluster, "watch_pods", None)
+    if stream_fn is None:
+        return watch(cluster, namespace, pod_names, mode, timeout=timeout)
+    watched = set(pod_names)
+    deadline = time.time() + timeout if timeout else None
+    stream = stream_fn(namespace)
+    try:
+        # initial sweep (events may predate subscription)
+        for name in list(watched):
+            p = cluster.get_pod(namespace, name)
+            if p is not None and _satisfied(p, mode):
+                watched.discard(name)
+        while watched:
+            if deadline and time.time() > deadline:
+                return False
+            ev = next(stream, StopIteration)
+            if ev is StopIteration:
+                return not watched
+            if ev is None:  # heartbeat
+                continue
+            if ev.name in watched and _satisfied(ev, mode):
+                watched.discard(ev.name)
+        return True
+    finally:
+        close = getattr(stream, "close", None)
+        if close:
+            close()
+
+
 def main(argv=None):
     """CLI parity with the reference watcher-loop binary
     (/root/reference/watcher-loop/app/server.go:38-64): env NAMESPACE,

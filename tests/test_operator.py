@@ -4,6 +4,8 @@ which never actually runs there because envtest has no kubelet. Here the
 FakeCluster's driveable kubelet makes the ladder testable:
 Pending -> Partitioning -> Partitioned -> Training -> Completed.
 """
+import time
+
 import pytest
 
 from dgl_operator_amd.operator_plane import (
@@ -242,6 +244,46 @@ def test_watcher_ready_and_finished():
     assert watcher.watch(c, "default", names, "finished", timeout=2)
     # launcher entries are skipped
     assert watcher.parse_watchfile(cm.data["leadfile"]) == []
+
+
+def test_watch_events_informer_mode():
+    from dgl_operator_amd.operator_plane.cluster import Pod
+    """Event-driven watch (the reference's informer + workqueue shape,
+    watcher-loop/controllers/controller.go:84-152): an already-satisfied
+    pod drains in the initial sweep; the rest drain on UPDATE events with
+    no polling; timeout works via heartbeats."""
+    import threading
+
+    c = FakeCluster()
+    for i in range(3):
+        c.create_pod(Pod(f"w-{i}", "default", owner="j"))
+    c.set_pod_phase("default", "w-0", PodPhase.RUNNING)  # pre-satisfied
+
+    def kubelet():
+        time.sleep(0.2)
+        c.set_pod_phase("default", "w-1", PodPhase.RUNNING)
+        time.sleep(0.1)
+        c.set_pod_phase("default", "w-2", PodPhase.RUNNING)
+
+    t = threading.Thread(target=kubelet, daemon=True)
+    t.start()
+    t0 = time.time()
+    assert watcher.watch_events(c, "default", ["w-0", "w-1", "w-2"],
+                                "ready", timeout=5)
+    assert time.time() - t0 < 3
+    t.join()
+    # finished-mode timeout (no events arrive)
+    assert not watcher.watch_events(c, "default", ["w-1"], "finished",
+                                    timeout=0.5)
+    # all subscriber queues unregistered
+    assert c._watchers == []
+    # clusters without a stream fall back to polling
+    class NoStream(FakeCluster):
+        watch_pods = None
+    ns = NoStream()
+    ns.create_pod(Pod("x", "default"))
+    ns.set_pod_phase("default", "x", PodPhase.RUNNING)
+    assert watcher.watch_events(ns, "default", ["x"], "ready", timeout=1)
 
 
 def test_evicted_launcher_is_retried():
