@@ -38,6 +38,35 @@ def test_bench_json_contract():
     assert "global_batch" in d["config"] and "parallelism" in d["config"]
 
 
+@pytest.mark.timeout(400)
+@pytest.mark.parametrize("halo", [True, False])
+def test_bench_two_rank_driver_contract(halo):
+    """The exact launch shape the driver uses for N>1 (torchrun, one rank
+    per 'GPU' — gloo/CPU here), both halo modes: whole-job aggregate value,
+    max-over-ranks time, dp2 parallelism string. De-risks the round-end
+    8-GPU SCALE run on the paths a single GPU cannot cover."""
+    args = [sys.executable, "-m", "torch.distributed.run", "--standalone",
+            "--local-addr", "127.0.0.1", "--nproc-per-node", "2",
+            os.path.join(REPO, "bench.py"), "--gpus", "2",
+            "--steps", "2", "--warmup", "1",
+            "--nodes", "20000", "--edges", "100000", "--batch", "200"]
+    if not halo:
+        args.append("--no-halo")
+    for attempt in range(2):  # absorbs rendezvous races
+        r = _run(args, timeout=360)
+        if r.returncode == 0:
+            break
+    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+    d = json.loads([l for l in r.stdout.splitlines()
+                    if l.startswith("{")][-1])
+    assert d["n_gpus"] == 2
+    assert d["config"]["global_batch"] == 400
+    assert d["config"]["parallelism"].startswith("dp2")
+    mode = "ghost-zone" if halo else "alltoallv"
+    assert mode in d["config"]["parallelism"]
+    assert d["value"] > 0
+
+
 @pytest.mark.timeout(240)
 @pytest.mark.parametrize("script,args", [
     ("examples/node_classification/train.py", ["--epochs", "3", "--feat", "32"]),
