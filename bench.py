@@ -65,6 +65,11 @@ def parse_args():
                         "(eager mode, CUDA events)")
     p.add_argument("--capture", action="store_true", default=None,
                    help="force hipGraph capture of the whole train step")
+    p.add_argument("--no-prefetch", dest="prefetch", action="store_false",
+                   default=True,
+                   help="disable side-stream sampling prefetch in eager "
+                        "mode (prefetch overlaps step k+1's sampling chain "
+                        "+ host syncs with step k's fwd/bwd/opt)")
     p.add_argument("--no-capture", dest="capture", action="store_false",
                    help="disable the capture attempt (eager stepping)")
     p.add_argument("--capture-dist", action="store_true",
@@ -309,6 +314,62 @@ def main():
                 phase_ms[k] += ev[i].elapsed_time(ev[i + 1])
         return sum(b.num_edges for b in blocks)
 
+    # -- eager-mode sampling prefetch: step k+1's sampling chain (and its
+    # per-hop host syncs) runs on a SIDE stream while step k's fwd/bwd/opt
+    # execute on the main stream. Comm/compute overlap per the CDNA guide;
+    # sampler state is safe because ALL sampling runs on the side stream
+    # (serialized there) and training never touches the workspace.
+    # ws>1 without halo would issue sampling alltoallv collectives on the
+    # side stream concurrently with the main-stream grad all-reduce —
+    # cross-stream collective interleaving is a deadlock hazard, so
+    # prefetch stays off there; halo-mode sampling is communication-free
+    prefetch_ok = (args.prefetch and device.type == "cuda"
+                   and not args.phase_timing
+                   and (ws == 1 or not args.no_halo))
+
+    def _make_prefetched_step(base_step_edges_fn):
+        side = torch.cuda.Stream()
+        box = {}
+
+        def record_all(blocks, y, stream):
+            for b in blocks:
+                for t in (b.csc_indptr, b.csc_indices, b.srcdata_nids):
+                    if t is not None:
+                        t.record_stream(stream)
+            y.record_stream(stream)
+
+        def sample_next(step):
+            with torch.cuda.stream(side):
+                seeds = next_seeds()
+                inp, out_nodes, blocks = dg.sample_blocks(
+                    seeds, fanouts, seed=step + 1)
+                y = dg.pull("label", out_nodes)
+                ev = torch.cuda.Event()
+                ev.record(side)
+                box[step] = (inp, blocks, y, ev)
+
+        def one_step_prefetched(step: int) -> int:
+            if step not in box:
+                sample_next(step)
+            inp, blocks, y, ev = box.pop(step)
+            cur = torch.cuda.current_stream()
+            cur.wait_event(ev)
+            # tensors were allocated on the side stream; pin their reuse
+            # to the consuming stream for the caching allocator
+            record_all(blocks, y, cur)
+            x = dg.pull_view("feat", inp)
+            logits = model(blocks, x)
+            loss = F.cross_entropy(logits, y)
+            opt.zero_grad(set_to_none=True)
+            loss.backward()
+            flat_allreduce_grads(model)
+            opt.step()
+            # overlap: sample step+1 while the enqueued fwd/bwd/opt run
+            sample_next(step + 1)
+            return sum(b.num_edges for b in blocks)
+
+        return one_step_prefetched
+
     # -- hipGraph-captured step (1-GPU): sampling, compaction, gather, fwd,
     # bwd and Adam replay as ONE graph launch. Worst-case shapes keep every
     # tensor static; actual sizes live on device
@@ -324,6 +385,12 @@ def main():
             print(f"# capture: disabled ({type(e).__name__}: {msg}); "
                   "eager stepping")
             opt = torch.optim.Adam(model.parameters(), lr=args.lr)
+
+    use_prefetch = prefetch_ok and not use_capture
+    if use_prefetch:
+        one_step = _make_prefetched_step(one_step)
+        if rank == 0:
+            print("# prefetch: enabled (side-stream sampling overlap)")
 
     # warmup
     for s in range(args.warmup):
@@ -401,7 +468,8 @@ def main():
                        else "alltoallv halo pulls)")
                 ),
                 "step_mode": ("hipGraph-captured" if use_capture
-                              else "eager"),
+                              else ("eager+prefetch" if use_prefetch
+                                    else "eager")),
             },
         }))
 
