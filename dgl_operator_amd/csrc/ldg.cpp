@@ -34,12 +34,6 @@ static inline uint64_t mix64(uint64_t x) {
 
 namespace {
 
-struct Caps {
-  double cap, tcap, ecap;
-  const bool* tm;
-  bool balance_edges;
-};
-
 // neighbor-count phase for one node: counts[P] over both directions
 static inline void count_neighbors(int64_t v, const int64_t* ip,
                                    const int64_t* ix, const int64_t* cip,
@@ -163,7 +157,7 @@ at::Tensor ldg_partition(at::Tensor indptr, at::Tensor indices,
   // a move is applied only while it keeps every balance cap.
   const int refine_passes = 3;
   for (int pass = 0; pass < refine_passes; ++pass) {
-    int64_t moved = 0, want = 0, blocked_cap = 0, blocked_ecap = 0;
+    int64_t moved = 0;
     for (int64_t s = 0; s < n; s += max_chunk) {
       const int64_t e = std::min(s + max_chunk, n);
       at::parallel_for(s, e, 512, [&](int64_t b0, int64_t b1) {
@@ -186,22 +180,17 @@ at::Tensor ldg_partition(at::Tensor indptr, at::Tensor indices,
           if (q == cur) continue;
           const int32_t gain = counts[q] - counts[cur];
           if (gain <= best_gain) continue;
-          ++want;
           // a move is balance-safe if the target stays under the cap OR
           // remains no heavier than the source was (the stream saturates
           // the caps, so a pure cap check would freeze refinement)
-          if (sizes[q] + 1 > (int64_t)cap && sizes[q] + 1 > sizes[cur]) {
-            ++blocked_cap;
+          if (sizes[q] + 1 > (int64_t)cap && sizes[q] + 1 > sizes[cur])
             continue;
-          }
           if (tm && tm[v] && tsizes[q] + 1.0 > tcap &&
               tsizes[q] + 1.0 > tsizes[cur])
             continue;
           if (balance_edges && esizes[q] + vdeg > ecap &&
-              esizes[q] + vdeg > esizes[cur]) {
-            ++blocked_ecap;
+              esizes[q] + vdeg > esizes[cur])
             continue;
-          }
           best_gain = gain;
           best = (int)q;
         }
@@ -221,7 +210,6 @@ at::Tensor ldg_partition(at::Tensor indptr, at::Tensor indices,
         }
       }
     }
-    printf("[ldg] refine pass %d moved %ld want %ld bcap %ld becap %ld\n", pass, (long)moved, (long)want, (long)blocked_cap, (long)blocked_ecap);
     if (moved == 0) break;
   }
   return out;
