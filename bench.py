@@ -388,12 +388,22 @@ def main():
 
     use_prefetch = prefetch_ok and not use_capture
     if use_prefetch:
-        one_step = _make_prefetched_step(one_step)
-        if rank == 0:
-            print("# prefetch: enabled (side-stream sampling overlap)")
+        base_step = one_step
+        one_step = _make_prefetched_step(base_step)
+        # prove the prefetched path on this box before relying on it
+        try:
+            one_step(0)
+            torch.cuda.synchronize()
+            if rank == 0:
+                print("# prefetch: enabled (side-stream sampling overlap)")
+        except Exception as e:  # noqa: BLE001
+            use_prefetch = False
+            one_step = base_step
+            msg = str(e).splitlines()[0] if str(e) else type(e).__name__
+            print(f"# prefetch: disabled ({type(e).__name__}: {msg})")
 
     # warmup
-    for s in range(args.warmup):
+    for s in range(1 if use_prefetch else 0, args.warmup):
         one_step(s)
 
     if args.profile and rank == 0:
