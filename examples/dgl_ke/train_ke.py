@@ -72,11 +72,23 @@ def _capture_ke_step(model, sampler, args, device):
         for _ in range(4):  # warm both corruption sides
             body(fill())
     th.cuda.current_stream().wait_stream(side)
+    # capture-mode ladder (same as bench.py): "global" forbids ANY thread's
+    # unsafe HIP calls during capture and backward() runs on autograd
+    # worker threads — thread_local/relaxed scope the check to this thread
     for nh in (False, True):
-        g = th.cuda.CUDAGraph()
-        with th.cuda.graph(g):
-            body(nh)
-        graphs[nh] = g
+        last = None
+        for mode in ("global", "thread_local", "relaxed"):
+            g = th.cuda.CUDAGraph()
+            try:
+                with th.cuda.graph(g, capture_error_mode=mode):
+                    body(nh)
+                graphs[nh] = g
+                break
+            except Exception as e:  # noqa: BLE001
+                last = e
+                th.cuda.synchronize()
+        if nh not in graphs:
+            raise last
     # sanity replay both sides
     for _ in range(2):
         nh = fill()
