@@ -192,6 +192,14 @@ def partition_graph(
             "part_graph": f"part{p}/graph.pt",
             "node_feats": f"part{p}/node_feat.pt",
         }
+        if g.edata:
+            # edge features of the part's in-edges, aligned with the
+            # (src_global, dst_global) order in graph.pt — the reference
+            # dispatches edge_feat.dgl per part the same way
+            # (dispatch.py:80-91)
+            efeats = {k: v[emask].clone() for k, v in g.edata.items()}
+            torch.save(efeats, os.path.join(pdir, "edge_feat.pt"))
+            parts[f"part-{p}"]["edge_feats"] = f"part{p}/edge_feat.pt"
 
     spec = PartitionSpec(
         name=name,
@@ -209,11 +217,16 @@ def partition_graph(
 def load_partition(
     json_path: str, part_id: int
 ) -> Tuple[dict, Dict[str, torch.Tensor], PartitionSpec]:
-    """Load one partition. Returns (graph payload, node feature dict, spec)."""
+    """Load one partition. Returns (graph payload, node feature dict, spec).
+    Edge features, when partitioned, ride in the graph payload under
+    ``edge_feats`` (rows aligned with src_global/dst_global)."""
     with open(json_path) as f:
         spec = PartitionSpec.from_json(json.load(f))
     base = os.path.dirname(json_path)
     entry = spec.parts[f"part-{part_id}"]
     gpart = torch.load(os.path.join(base, entry["part_graph"]), weights_only=True)
     feats = torch.load(os.path.join(base, entry["node_feats"]), weights_only=True)
+    if "edge_feats" in entry:
+        gpart["edge_feats"] = torch.load(
+            os.path.join(base, entry["edge_feats"]), weights_only=True)
     return gpart, feats, spec

@@ -198,3 +198,29 @@ def test_kubexec_fabric_command_contract(monkeypatch):
     assert calls[0].startswith("/etc/dgl/kubexec.sh job-worker-0 A=1 echo hi")
     assert calls[1] == ("/opt/kube/kubectl cp /tmp/x job-launcher:/w/x"
                        " -c watcher-loop-partitioner")
+
+
+def test_partition_edge_features(tmp_path):
+    """g.edata is partitioned alongside the structure (reference ships
+    edge_feat.dgl per part, dispatch.py:80-91): per-part rows align with
+    the (src_global, dst_global) edge order."""
+    from dgl_operator_amd.graph.partition import load_partition
+
+    g = rmat_graph(50, 300, num_feats=4, seed=6)
+    g.edata["w"] = torch.arange(g.num_edges, dtype=torch.float32)
+    partition_graph(g, "ew", 2, str(tmp_path), algorithm="range")
+    total = 0
+    src, dst = g.edges()
+    for p in range(2):
+        gpart, feats, spec = load_partition(str(tmp_path / "ew.json"), p)
+        assert "edge_feats" in gpart
+        w = gpart["edge_feats"]["w"]
+        assert w.numel() == gpart["src_global"].numel()
+        total += w.numel()
+        # edge weights identify original edges: endpoints must match
+        # (range partition => new ids == old ids)
+        for k in range(min(5, w.numel())):
+            e = int(w[k])
+            assert int(src[e]) == int(gpart["src_global"][k])
+            assert int(dst[e]) == int(gpart["dst_global"][k])
+    assert total == g.num_edges
