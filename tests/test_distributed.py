@@ -319,3 +319,22 @@ def _uneven_worker(rank, world):
 
 def test_uneven_shards_no_deadlock():
     _run_workers(_uneven_worker)
+
+
+def _many_rank_worker(rank, world):
+    """pull/push/sampling/halo at world>2: multi-segment alltoallv reorder
+    paths that symmetric 2-rank tests cannot distinguish (a swapped segment
+    is its own inverse at world=2)."""
+    _pull_worker(rank, world)
+    _push_worker(rank, world)
+    _sample_worker(rank, world)
+    _halo_worker(rank, world)
+
+
+def test_distributed_world4():
+    _run_workers(_many_rank_worker, world=4)
+
+
+def test_distributed_world3_uneven_shards():
+    # 200 nodes / 3 parts -> 66/67/67: uneven owned ranges through every path
+    _run_workers(_many_rank_worker, world=3)
