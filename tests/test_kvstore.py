@@ -205,3 +205,14 @@ def test_kge_eval_corrupt_head():
     res = evaluate_kge(m, h, r, t, corrupt="head")
     assert 0.0 < res["MRR"] <= 1.0
     assert 1.0 <= res["MR"] <= 50.0
+
+
+def _kv_kge_many_rank_worker(rank, world):
+    _kv_worker(rank, world)
+    _kge_worker(rank, world)
+
+
+def test_kvstore_and_kge_world4():
+    """Sharded embedding + KGE train at world=4: multi-segment pull/push
+    reorder paths (2-rank swaps are self-inverse and can hide bugs)."""
+    _run_workers(_kv_kge_many_rank_worker, world=4)
