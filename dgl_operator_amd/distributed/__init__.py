@@ -3,5 +3,6 @@ from .partition_book import PartitionBook
 from .dist_graph import DistGraph
 from .kvstore import ShardedEmbedding
 from .kge import DistKGEModel, KGEdgeSampler
+from .dist_tensor import DistTensor, DistNodeDataLoader
 
 __all__ = ["comm", "PartitionBook", "DistGraph", "ShardedEmbedding", "DistKGEModel", "KGEdgeSampler"]

@@ -138,27 +138,22 @@ def main():
 
     train_nids = dg.node_split("train_mask")
 
-    gen = torch.Generator(device=device)
-    gen.manual_seed(1234 + rank)
-    # ranks own different train-node counts; every rank must run the SAME
-    # number of steps or the gradient all-reduce deadlocks — reduce to MIN
-    steps_per_epoch = max(1, train_nids.numel() // args.batch_size)
-    if ws > 1:
-        # RCCL (backend 'nccl') rejects CPU tensors — reduce on the device
-        spe = torch.tensor([steps_per_epoch], device=device)
-        dist.all_reduce(spe, op=dist.ReduceOp.MIN)
-        steps_per_epoch = max(1, int(spe[0]))
+    # DistDataLoader + NeighborSampler analog (reference train_dist.py:
+    # 52-70,215): per-epoch shuffled slices of the owned train nodes; the
+    # loader all-reduces steps_per_epoch to the MIN so the gradient
+    # all-reduce (and non-halo sampling collectives) cannot desynchronize
+    from dgl_operator_amd.distributed import DistNodeDataLoader
+
+    loader = DistNodeDataLoader(dg, train_nids, fanouts, args.batch_size,
+                                seed=1234, epoch=start_epoch)
 
     for epoch in range(start_epoch, args.num_epochs):
         t_epoch = time.time()
-        for step in range(steps_per_epoch):
+        step = -1
+        for inp, out_nodes, blocks in loader:
+            step += 1
             tic = time.time()
-            sel = torch.randint(0, train_nids.numel(), (args.batch_size,),
-                                generator=gen, device=device)
-            seeds = torch.unique(train_nids[sel])
-            inp, out_nodes, blocks = dg.sample_blocks(
-                seeds, fanouts, seed=epoch * 100000 + step
-            )
+            seeds = out_nodes
             x = dg.pull_view("feat", inp)
             y = dg.pull("label", out_nodes)
             t_sample = time.time()

@@ -338,3 +338,62 @@ def test_distributed_world4():
 def test_distributed_world3_uneven_shards():
     # 200 nodes / 3 parts -> 66/67/67: uneven owned ranges through every path
     _run_workers(_many_rank_worker, world=3)
+
+
+def _dist_tensor_worker(rank, world):
+    from dgl_operator_amd.distributed.dist_tensor import DistTensor
+
+    t = DistTensor((100, 4))
+    # overwrite by global id from every rank (disjoint ids per rank)
+    ids = torch.arange(rank * 10, rank * 10 + 10)
+    t[ids] = torch.full((10, 4), float(rank + 1))
+    dist.barrier()
+    # read back arbitrary global ids (collective: all ranks, same count)
+    probe = torch.tensor([0, 5, 15, 99, (world - 1) * 10])
+    rows = t[probe]
+    for i, gid in enumerate(probe.tolist()):
+        owner_write = gid // 10
+        expect = float(owner_write + 1) if owner_write < world and \
+            gid < world * 10 else 0.0
+        assert torch.allclose(rows[i], torch.full((4,), expect)), (gid, rows[i])
+    # accumulating write: every rank adds 1.0 into id 50
+    t.index_add_(torch.tensor([50]), torch.ones(1, 4))
+    dist.barrier()
+    got = t[torch.tensor([50])]
+    base = 6.0 if 50 < world * 10 else 0.0  # overwritten by rank 5 at world>5
+    assert torch.allclose(got[0], torch.full((4,), base + world)), got
+
+
+def test_dist_tensor():
+    _run_workers(_dist_tensor_worker)
+    _run_workers(_dist_tensor_worker, world=4)
+
+
+def _loader_worker(rank, world):
+    from dgl_operator_amd.distributed.dist_tensor import DistNodeDataLoader
+
+    g, dg = _make_shard(rank, world)
+    nids = dg.owned_nodes()
+    loader = DistNodeDataLoader(dg, nids, [3, 5], batch_size=16, seed=2)
+    # same step count on every rank
+    n = torch.tensor([len(loader)])
+    lo, hi = n.clone(), n.clone()
+    dist.all_reduce(lo, op=dist.ReduceOp.MIN)
+    dist.all_reduce(hi, op=dist.ReduceOp.MAX)
+    assert torch.equal(lo, hi)
+    seen = 0
+    for inp, seeds, blocks in loader:
+        assert seeds.numel() <= 16
+        assert len(blocks) == 2
+        assert blocks[-1].num_dst_nodes == seeds.numel()
+        # seeds are owned nodes
+        assert (seeds >= dg.lo).all() and (seeds < dg.hi).all()
+        seen += 1
+    assert seen == len(loader)
+    # epoch advances -> different first batch
+    first = next(iter(loader))[1]
+    assert loader.epoch == 2 or not torch.equal(first, seeds)
+
+
+def test_dist_node_dataloader():
+    _run_workers(_loader_worker)
