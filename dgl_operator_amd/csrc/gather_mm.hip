@@ -108,7 +108,8 @@ at::Tensor gather_mm(at::Tensor feat, at::Tensor rows, at::Tensor weight,
   const size_t lds_bytes =
       (GM_ROWS * (Kp + 1) + Kp * GM_N) * sizeof(float);
   // HIP caps dynamic LDS at 64 KiB unless the max-dynamic-shared attribute
-  // is raised; K <= ~230 fits comfortably under the default cap
+  // is raised; 80*Kp + 64 floats <= 16384 => Kp <= 204 (callers guard at
+  // K <= 204 and fall back to gather_rows + rocBLAS GEMM beyond)
   TORCH_CHECK(lds_bytes <= 64 * 1024, "gather_mm: K too large for LDS");
   const int grid = grid_for(ceil_div(M, GM_ROWS) * 256, 256);
   auto stream = c10::hip::getCurrentHIPStream().stream();

@@ -84,7 +84,7 @@ class SAGEConv(nn.Module):
                 x.feat.is_cuda
                 and x.feat.dtype == torch.float32
                 and self.fc_neigh.out_features <= 16
-                and x.feat.shape[1] <= 230  # gather_mm LDS envelope
+                and x.feat.shape[1] <= 204  # gather_mm LDS envelope (64*(Kp+1)+Kp*16 floats <= 64 KiB => Kp <= 204)
             )
             if not fusable:
                 x = x.materialize()

@@ -23,7 +23,7 @@ class _GatherMM(torch.autograd.Function):
     def forward(ctx, feat, rows, weight, bias):
         ctx.save_for_backward(feat, rows)
         ctx.has_bias = bias is not None
-        if (feat.is_cuda and weight.shape[1] <= 16 and feat.shape[1] <= 230
+        if (feat.is_cuda and weight.shape[1] <= 16 and feat.shape[1] <= 204
                 and feat.dtype == torch.float32 and backend.has_extension()):
             ext = backend.ext_for(feat)
             return ext.gather_mm(feat, rows, weight.contiguous(), bias)
