@@ -145,7 +145,9 @@ def main():
 
     rank, ws = comm.init_from_env()
     if torch.cuda.is_available():
-        device = torch.device(f"cuda:{int(os.environ.get('LOCAL_RANK', 0))}")
+        device = torch.device(
+            f"cuda:{int(os.environ.get('LOCAL_RANK', 0)) % torch.cuda.device_count()}"
+        )  # rank % num_gpus, reference train_dist.py:282-285
         torch.cuda.set_device(device)
     else:
         device = torch.device("cpu")

@@ -73,7 +73,9 @@ def main():
     rank, ws = comm.init_from_env()
     force_cpu = args.num_gpus is not None and args.num_gpus == -1
     if torch.cuda.is_available() and not force_cpu:
-        device = torch.device(f"cuda:{int(os.environ.get('LOCAL_RANK', 0))}")
+        device = torch.device(
+            f"cuda:{int(os.environ.get('LOCAL_RANK', 0)) % torch.cuda.device_count()}"
+        )  # rank % num_gpus, reference train_dist.py:282-285
         torch.cuda.set_device(device)
     else:
         device = torch.device("cpu")
