@@ -155,6 +155,17 @@ def run_launcher(args):
         raise SystemExit(
             f"[dglrun] --num-workers {args.num_workers} but hostfile has "
             f"{len(hosts)} workers")
+    if args.num_trainers:
+        # one partition per RANK: --num-trainers overrides ranks-per-pod, so
+        # dispatch must ship partitions per the same count (keeping hostfile
+        # slots for dispatch while torchrun launches num_trainers ranks
+        # would desync partition ownership)
+        for h in hosts:
+            if h.slots != args.num_trainers:
+                print(f"[dglrun] note: hostfile slots={h.slots} overridden "
+                      f"by --num-trainers {args.num_trainers} for {h.pod}",
+                      flush=True)
+            h.slots = args.num_trainers
     dataset = os.path.join(args.workspace, "dataset")
     with phase("Phase 3/5 dispatch"):
         if args.dispatch_entry_point:
