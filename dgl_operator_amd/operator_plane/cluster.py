@@ -388,6 +388,59 @@ class KubectlCluster(Cluster):
         finally:
             proc.terminate()
 
+    # -- leader election (coordination.k8s.io Lease; main.go:73-80 parity) --
+    def try_acquire_lease(self, namespace: str, name: str, identity: str,
+                          ttl_seconds: int = 15) -> bool:
+        """Acquire or renew the leader Lease. Returns True while this
+        identity holds it. Same semantics as controller-runtime's
+        leasing resource lock: a lease held by another identity is only
+        taken over once renewTime is older than the ttl."""
+        import datetime
+
+        now = datetime.datetime.now(datetime.timezone.utc)
+        now_s = now.strftime("%Y-%m-%dT%H:%M:%S.%f")[:-3] + "Z"
+        d = self._get_json(["get", "lease", name, "-n", namespace])
+        if d is None:
+            try:
+                self._create({
+                    "apiVersion": "coordination.k8s.io/v1",
+                    "kind": "Lease",
+                    "metadata": {"name": name, "namespace": namespace},
+                    "spec": {
+                        "holderIdentity": identity,
+                        "leaseDurationSeconds": ttl_seconds,
+                        "acquireTime": now_s,
+                        "renewTime": now_s,
+                    },
+                })
+                return True
+            except RuntimeError:
+                return False  # lost the creation race
+        spec = d.get("spec", {})
+        holder = spec.get("holderIdentity")
+        if holder and holder != identity:
+            renew = spec.get("renewTime") or spec.get("acquireTime")
+            if renew:
+                try:
+                    t = datetime.datetime.strptime(
+                        renew.replace("Z", "+0000"),
+                        "%Y-%m-%dT%H:%M:%S.%f%z")
+                except ValueError:
+                    t = datetime.datetime.strptime(
+                        renew.replace("Z", "+0000"), "%Y-%m-%dT%H:%M:%S%z")
+                if (now - t).total_seconds() < spec.get(
+                        "leaseDurationSeconds", ttl_seconds):
+                    return False  # current leader is live
+        # take over / renew
+        patch = json.dumps({"spec": {
+            "holderIdentity": identity,
+            "leaseDurationSeconds": ttl_seconds,
+            "renewTime": now_s,
+        }})
+        r = self._run(["patch", "lease", name, "-n", namespace,
+                       "--type=merge", "-p", patch])
+        return r.returncode == 0
+
     # -- DGLJob CRs (the Manager watch surface) -----------------------------
     def list_dgljob_manifests(self, namespace: "str | None" = None):
         """Raw DGLJob CR dicts (all namespaces unless one is given)."""

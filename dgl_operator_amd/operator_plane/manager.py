@@ -12,6 +12,7 @@ real deployment.
 from __future__ import annotations
 
 import argparse
+import os
 import threading
 import time
 from http.server import BaseHTTPRequestHandler, HTTPServer
@@ -54,11 +55,17 @@ class _HealthHandler(BaseHTTPRequestHandler):
         pass
 
 
+LEADER_ELECTION_ID = "9007c5fc.qihoo.net"  # main.go:76
+
+
 class Manager:
     def __init__(self, cluster: Optional[Cluster] = None,
                  reconcile_interval: float = 0.5,
                  watcher_loop_image: str = "watcher-loop",
-                 kubectl_download_image: str = "kubectl-download"):
+                 kubectl_download_image: str = "kubectl-download",
+                 leader_elect: bool = False,
+                 election_namespace: str = "dgl-operator",
+                 lease_ttl: float = 15.0):
         self.cluster = cluster or FakeCluster()
         self.reconciler = DGLJobReconciler(
             self.cluster, watcher_loop_image, kubectl_download_image
@@ -70,6 +77,29 @@ class Manager:
         # real-cluster mode: poll DGLJob CRs + write status back
         self.watch_crs = hasattr(self.cluster, "list_dgljob_manifests")
         self._written_statuses: Dict[str, dict] = {}
+        # leader election (reference main.go:73-80): only the Lease holder
+        # reconciles; replicas keep contending every loop and take over
+        # when the holder stops renewing for a ttl
+        self.leader_elect = (leader_elect
+                             and hasattr(self.cluster, "try_acquire_lease"))
+        self.election_namespace = election_namespace
+        self.lease_ttl = lease_ttl
+        import socket
+
+        self.identity = f"{socket.gethostname()}_{os.getpid()}"
+        self.is_leader = not self.leader_elect
+
+    def _ensure_leadership(self) -> bool:
+        if not self.leader_elect:
+            return True
+        was = self.is_leader
+        self.is_leader = self.cluster.try_acquire_lease(
+            self.election_namespace, LEADER_ELECTION_ID, self.identity,
+            int(self.lease_ttl))
+        if self.is_leader != was:
+            print(f"[manager] leadership {'acquired' if self.is_leader else 'lost'}"
+                  f" ({self.identity})", flush=True)
+        return self.is_leader
 
     # -- job API (the CRD surface) -----------------------------------------
     def submit(self, manifest) -> DGLJob:
@@ -128,6 +158,8 @@ class Manager:
 
     # -- reconcile loop ----------------------------------------------------
     def reconcile_once(self):
+        if not self._ensure_leadership():
+            return
         if self.watch_crs:
             self.sync_from_cluster()
         for key, job in list(self.jobs.items()):
@@ -188,8 +220,10 @@ def main(argv=None):
     p.add_argument("--health-probe-bind-address", default=":8081")
     p.add_argument("--reconcile-interval", type=float, default=0.5)
     p.add_argument("--leader-elect", action="store_true",
-                   help="accepted for flag parity with the reference manager; "
-                        "single-replica deployments need no election")
+                   help="contend for the leader Lease "
+                        f"(id {LEADER_ELECTION_ID}); only the holder "
+                        "reconciles — multi-replica deployments")
+    p.add_argument("--election-namespace", default="dgl-operator")
     p.add_argument("--watcher-loop-image", default="watcher-loop")
     p.add_argument("--kubectl-download-image", default="kubectl-download")
     p.add_argument("--job", action="append", default=[],
@@ -219,7 +253,9 @@ def main(argv=None):
     mgr = Manager(cluster=cluster,
                   reconcile_interval=args.reconcile_interval,
                   watcher_loop_image=args.watcher_loop_image,
-                  kubectl_download_image=args.kubectl_download_image)
+                  kubectl_download_image=args.kubectl_download_image,
+                  leader_elect=args.leader_elect,
+                  election_namespace=args.election_namespace)
     mode = "cluster (DGLJob CR watch)" if mgr.watch_crs else "local --job files"
     print(f"[manager] mode: {mode}", flush=True)
     for path in args.job:

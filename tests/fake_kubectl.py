@@ -39,12 +39,14 @@ SINGULAR = {
     "roles": "role", "role": "role",
     "rolebindings": "rolebinding", "rolebinding": "rolebinding",
     "dgljobs": "dgljob", "dgljob": "dgljob",
+    "leases": "lease", "lease": "lease",
 }
 
 KIND_NAME = {
     "pod": "Pod", "configmap": "ConfigMap", "service": "Service",
     "serviceaccount": "ServiceAccount", "role": "Role",
     "rolebinding": "RoleBinding", "dgljob": "DGLJob",
+    "lease": "Lease",
 }
 
 

@@ -273,3 +273,41 @@ def test_job_from_manifest_uid_and_deletion():
     job = job_from_manifest(m)
     assert job.uid == "u-1"
     assert job.deletion_timestamp is not None
+
+
+def test_leader_election(kube):
+    """Only the Lease holder reconciles (main.go:73-80 parity); a standby
+    takes over once the holder stops renewing for a ttl."""
+    import time
+
+    c, state = kube
+    manifest = _submit_cr(
+        c, state,
+        os.path.join(HERE, "..", "examples", "v1alpha1", "GraphSAGE.yaml"))
+    ns, name = (manifest["metadata"]["namespace"],
+                manifest["metadata"]["name"])
+
+    a = Manager(cluster=c, leader_elect=True, lease_ttl=1.0)
+    b = Manager(cluster=c, leader_elect=True, lease_ttl=1.0)
+    b.identity = a.identity + "-standby"
+
+    a.reconcile_once()
+    assert a.is_leader
+    assert read_obj(state, "pod", ns, f"{name}-launcher") is not None
+    lease = read_obj(state, "lease", "dgl-operator", "9007c5fc.qihoo.net")
+    assert lease["spec"]["holderIdentity"] == a.identity
+
+    # the standby must NOT reconcile while a's lease is fresh
+    os.unlink(state / f"pod__{ns}__{name}-launcher.json")
+    b.reconcile_once()
+    assert not b.is_leader
+    assert read_obj(state, "pod", ns, f"{name}-launcher") is None
+
+    # holder stops renewing -> standby takes over after the ttl
+    time.sleep(1.2)
+    b.reconcile_once()
+    assert b.is_leader
+    assert read_obj(state, "pod", ns, f"{name}-launcher") is not None
+    # a contends again but the lease is freshly held by b
+    a.reconcile_once()
+    assert not a.is_leader
