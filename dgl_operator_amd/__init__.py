@@ -29,7 +29,7 @@ tools         dglrun workflow CLI, launch/dispatch/revise_hostfile
               (reference: python/dglrun/).
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 try:  # compute plane needs torch; the control-plane-only manager image
     from . import graph  # noqa: F401
