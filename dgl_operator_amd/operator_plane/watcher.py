@@ -145,8 +145,10 @@ def main(argv=None):
         names = parse_watchfile(f.read())
     print(f"[watcher-loop] waiting for {len(names)} pods to be {args.mode}",
           flush=True)
-    ok = watch(KubectlCluster(), args.namespace, names, args.mode,
-               timeout=args.timeout)
+    # event-driven (informer-style) against the real apiserver; watch()
+    # remains the fallback for clusters without a watch stream
+    ok = watch_events(KubectlCluster(), args.namespace, names, args.mode,
+                      timeout=args.timeout)
     sys.exit(0 if ok else 1)
 
 
