@@ -69,6 +69,8 @@ class DistKGEModel:
         lr: float,
         neg_head: bool = False,
         adversarial_temperature: float = 1.0,
+        regularization_coef: float = 0.0,
+        regularization_norm: int = 3,
     ) -> float:
         B = heads.numel()
         num_chunk, n_neg = neg_entities.shape
@@ -90,6 +92,14 @@ class DistKGEModel:
         nc = n.reshape(num_chunk, n_neg, dim)
         neg = self.score.neg(hc, rc, nc, neg_head=neg_head)
         loss = kge_loss(pos, neg, adversarial_temperature)
+        if regularization_coef:
+            # DGL-KE's Lp regularization of the batch's entity embeddings
+            # (the reference runs dglke_dist_train with
+            # --regularization_coef 1e-9, dglkerun:301; norm p=3 is the
+            # dglke default). Gradients flow into the same push.
+            p = regularization_norm
+            loss = loss + regularization_coef * (
+                rows.abs().pow(p).sum() / max(rows.shape[0], 1))
         loss.backward()
 
         with torch.no_grad():

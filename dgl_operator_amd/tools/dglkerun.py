@@ -55,6 +55,9 @@ def build_parser():
     p.add_argument("--batch-size", type=int, default=1024)
     p.add_argument("--neg-sample-size", type=int, default=256)
     p.add_argument("--max-step", type=int, default=1000)
+    p.add_argument("--regularization-coef", type=float, default=1e-9,
+                   help="forwarded to the trainer "
+                        "(reference dglkerun:301 fixed value)")
     p.add_argument("--save-path", default="ckpts")
     p.add_argument("--no-save-emb", action="store_true")
     p.add_argument("--ignore-partition", action="store_true")
@@ -172,7 +175,8 @@ def main(argv=None):
             f"--gamma {args.gamma} --lr {args.lr} "
             f"--batch-size {args.batch_size} "
             f"--neg-sample-size {args.neg_sample_size} "
-            f"--max-step {args.max_step} --save-path {args.save_path}"
+            f"--max-step {args.max_step} --save-path {args.save_path} "
+            f"--regularization-coef {args.regularization_coef}"
             + (" --no-save-emb" if args.no_save_emb else "")
         )
         if have_partition:

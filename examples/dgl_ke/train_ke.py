@@ -49,6 +49,10 @@ def _capture_ke_step(model, sampler, args, device):
         nc = n.reshape(C, args.neg_sample_size, -1)
         neg = model.score.neg(hc, rc, nc, neg_head=neg_head)
         loss = kge_loss(pos, neg, args.adversarial_temperature)
+        if args.regularization_coef:
+            loss = loss + args.regularization_coef * (
+                rows.abs().pow(args.regularization_norm).sum()
+                / max(rows.shape[0], 1))
         loss.backward()
         with th.no_grad():
             model.entities.push_grad(ent_ids, rows.grad, args.lr)
@@ -118,6 +122,12 @@ def main():
                    default=1.0, help="self-adversarial sampling temperature "
                                      "(dglke -adv is always on here)")
     p.add_argument("--max-step", type=int, default=1000)
+    p.add_argument("--regularization-coef", "--regularization_coef",
+                   type=float, default=0.0,
+                   help="Lp regularization of batch entity embeddings "
+                        "(reference dglkerun passes 1e-9)")
+    p.add_argument("--regularization-norm", "--regularization_norm",
+                   type=int, default=3)
     p.add_argument("--log-interval", type=int, default=100)
     p.add_argument("--num-entities", type=int, default=1_000_000)
     p.add_argument("--num-relations", type=int, default=1000)
@@ -218,6 +228,8 @@ def main():
                 hh, rr, tt, negs, args.chunk_size, args.lr,
                 neg_head=neg_head,
                 adversarial_temperature=args.adversarial_temperature,
+                regularization_coef=args.regularization_coef,
+                regularization_norm=args.regularization_norm,
             )
         if step % args.log_interval == 0 and rank == 0:
             if loss is None:

@@ -216,3 +216,36 @@ def test_kvstore_and_kge_world4():
     """Sharded embedding + KGE train at world=4: multi-segment pull/push
     reorder paths (2-rank swaps are self-inverse and can hide bugs)."""
     _run_workers(_kv_kge_many_rank_worker, world=4)
+
+
+def test_kge_regularization_term():
+    """--regularization_coef parity (reference dglkerun:301 passes 1e-9):
+    the Lp term changes the loss and pushes shrinkage gradients into the
+    batch embeddings."""
+    from dgl_operator_amd.distributed import DistKGEModel
+
+    torch.manual_seed(0)
+    model_a = DistKGEModel(50, 4, hidden_dim=8, score_func="TransE_l2",
+                           gamma=5.0, seed=3)
+    model_b = DistKGEModel(50, 4, hidden_dim=8, score_func="TransE_l2",
+                           gamma=5.0, seed=3)
+    h = torch.tensor([1, 2, 3, 4])
+    r = torch.tensor([0, 1, 2, 3])
+    t = torch.tensor([5, 6, 7, 8])
+    negs = torch.randint(0, 50, (1, 4))
+    la = model_a.train_step(h, r, t, negs, chunk_size=4, lr=0.0)
+    lb = model_b.train_step(h, r, t, negs, chunk_size=4, lr=0.0,
+                            regularization_coef=10.0)
+    # reg adds coef * mean_row sum |e|^3 of the pulled rows
+    rows = torch.cat([model_a.entities.local[i] for i in
+                      torch.cat([h, t, negs.reshape(-1)])]).view(-1, 8)
+    expect = 10.0 * rows.abs().pow(3).sum().item() / rows.shape[0]
+    assert abs((lb - la) - expect) / max(expect, 1e-9) < 1e-4
+    # with lr > 0 the reg term shrinks embeddings toward zero
+    big = DistKGEModel(50, 4, hidden_dim=8, score_func="TransE_l2",
+                       gamma=5.0, seed=3)
+    norm0 = float(big.entities.local[h].abs().sum())
+    for _ in range(5):
+        big.train_step(h, r, t, negs, chunk_size=4, lr=0.5,
+                       regularization_coef=100.0)
+    assert float(big.entities.local[h].abs().sum()) < norm0
