@@ -58,6 +58,11 @@ def build_parser():
     p.add_argument("--regularization-coef", type=float, default=1e-9,
                    help="forwarded to the trainer "
                         "(reference dglkerun:301 fixed value)")
+    p.add_argument("--test", action="store_true",
+                   help="evaluate after training (reference dglkerun:300 "
+                        "--test; maps to the trainer's --eval)")
+    p.add_argument("--batch-size-eval", type=int, default=1024)
+    p.add_argument("--log-interval", type=int, default=100)
     p.add_argument("--save-path", default="ckpts")
     p.add_argument("--no-save-emb", action="store_true")
     p.add_argument("--ignore-partition", action="store_true")
@@ -176,7 +181,10 @@ def main(argv=None):
             f"--batch-size {args.batch_size} "
             f"--neg-sample-size {args.neg_sample_size} "
             f"--max-step {args.max_step} --save-path {args.save_path} "
-            f"--regularization-coef {args.regularization_coef}"
+            f"--regularization-coef {args.regularization_coef} "
+            f"--log-interval {args.log_interval} "
+            f"--batch-size-eval {args.batch_size_eval}"
+            + (" --eval" if args.test else "")
             + (" --no-save-emb" if args.no_save_emb else "")
         )
         if have_partition:

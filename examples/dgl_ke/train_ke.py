@@ -149,6 +149,8 @@ def main():
     p.add_argument("--eval", action="store_true",
                    help="report raw MRR/MR/Hits@K on held-out triples")
     p.add_argument("--num-eval", type=int, default=500)
+    p.add_argument("--batch-size-eval", "--batch_size_eval", type=int,
+                   default=128, help="eval scoring batch (dglke parity)")
     args = p.parse_args()
 
     from dgl_operator_amd.distributed import DistKGEModel, KGEdgeSampler, comm
@@ -285,7 +287,8 @@ def main():
                 dist.broadcast(buf, src=0)
                 eh, er, et = buf[0], buf[1], buf[2]
         m = evaluate_kge(model, eh[: args.num_eval], er[: args.num_eval],
-                         et[: args.num_eval])
+                         et[: args.num_eval],
+                         batch_size=args.batch_size_eval)
         if rank == 0:
             print("eval:", {k: round(v, 4) for k, v in m.items()}, flush=True)
     if args.save_path and not args.no_save_emb:

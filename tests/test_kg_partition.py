@@ -151,13 +151,13 @@ def test_dglkerun_five_phase_local(tmp_path):
         "--num-partitions", "2",
         "--model", "TransE_l2", "--hidden-dim", "16",
         "--batch-size", "64", "--neg-sample-size", "8",
-        "--max-step", "20",
+        "--max-step", "20", "--log-interval", "10",
         "--workspace", "ws",
         "--hostfile", str(hostfile), "--leadfile", str(leadfile),
         "--master-port", str(port),
         "--train-entry-point",
         os.path.join(REPO, "examples", "dgl_ke", "train_ke.py")
-        + " --chunk-size 16 --log-interval 10 --no-capture",
+        + " --chunk-size 16 --no-capture",
     ]
     e = dict(env)
     e["DGL_OPERATOR_PHASE_ENV"] = "Partitioner"
