@@ -174,13 +174,22 @@ def _build_captured_step(args, dg, model, opt, device, fanouts, next_seeds,
             torch.cuda.synchronize()
     if graph is None:
         raise last_err
-    # sanity: two replays must accumulate edges and keep weights finite
+    # sanity: replays must count edges, run backward (nonzero grads), step
+    # the optimizer (weights move) and keep everything finite — a captured
+    # graph that silently skipped work would otherwise bench an invalid step
     edge_accum.zero_()
+    before = [p_.detach().clone() for p_ in model.parameters()]
     for s in range(2):
         fill_seeds(90_000 + s)
         graph.replay()
     torch.cuda.synchronize()
     assert float(edge_accum[0]) > 0, "captured step counted no edges"
+    grad_mag = sum(float(p_.grad.abs().sum()) for p_ in model.parameters()
+                   if p_.grad is not None)
+    assert grad_mag > 0, "captured step produced no gradients"
+    moved = any(not torch.equal(b, p_.detach())
+                for b, p_ in zip(before, model.parameters()))
+    assert moved, "captured step did not update weights"
     for p_ in model.parameters():
         assert bool(torch.isfinite(p_).all()), "non-finite weights"
     if rank == 0:
