@@ -93,13 +93,18 @@ def _capture_ke_step(model, sampler, args, device):
                 th.cuda.synchronize()
         if nh not in graphs:
             raise last
-    # sanity replay both sides
+    # sanity replay both sides: losses finite AND the sharded embeddings
+    # actually moved (the captured push+Adagrad ran — a graph that skipped
+    # the update would bench an invalid step)
+    ent_before = model.entities.local.detach().clone()
     for _ in range(2):
         nh = fill()
         graphs[nh].replay()
     th.cuda.synchronize()
     for nh in (False, True):
         assert bool(th.isfinite(losses[nh])), "non-finite captured loss"
+    assert not th.equal(ent_before, model.entities.local), \
+        "captured KE step did not update the entity shard"
 
     def replay(step):
         nh = fill()
