@@ -169,9 +169,14 @@ def run_launcher(args):
     dataset = os.path.join(args.workspace, "dataset")
     with phase("Phase 3/5 dispatch"):
         if args.dispatch_entry_point:
+            # custom dispatch scripts are most likely written against the
+            # reference contract (dglrun:182-188) — pass those spellings;
+            # the built-in dispatch CLI accepts them too
+            part_cfg = os.path.join(dataset, f"{args.graph_name}.json")
             _run(f"python {args.dispatch_entry_point} "
-                 f"--dataset-dir {dataset} --graph-name {args.graph_name} "
-                 f"--hostfile {args.hostfile} --workspace {args.workspace}")
+                 f"--workspace {args.workspace} "
+                 f"--rel_data_path dataset --rel_workload_path workload "
+                 f"--part_config {part_cfg} --ip_config {args.hostfile}")
         else:
             dispatch_partitions(dataset, args.graph_name, hosts,
                                 workspace=args.workspace)

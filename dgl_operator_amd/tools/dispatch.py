@@ -71,15 +71,32 @@ def dispatch_partitions(
 def main(argv=None):
     import argparse
 
+    # accepts BOTH this repo's spellings and the reference dispatch.py
+    # invocation contract (dglrun:182-188: --workspace --rel_data_path
+    # --rel_workload_path --part_config --ip_config)
     p = argparse.ArgumentParser()
-    p.add_argument("--dataset-dir", required=True)
-    p.add_argument("--graph-name", required=True)
-    p.add_argument("--hostfile", default="/etc/dgl/hostfile")
+    p.add_argument("--dataset-dir", default=None)
+    p.add_argument("--graph-name", default=None)
+    p.add_argument("--hostfile", "--ip_config", "--ip-config",
+                   default="/etc/dgl/hostfile")
     p.add_argument("--workspace", default="/dgl_workspace")
+    p.add_argument("--part_config", "--part-config", default=None,
+                   help="reference spelling: path to <dataset>/<graph>.json")
+    p.add_argument("--rel_data_path", "--rel-data-path", default="dataset")
+    p.add_argument("--rel_workload_path", "--rel-workload-path",
+                   default=WORKLOAD_DIR)
     args = p.parse_args(argv)
+    dataset_dir, graph_name = args.dataset_dir, args.graph_name
+    if args.part_config:
+        dataset_dir = dataset_dir or os.path.dirname(args.part_config)
+        graph_name = graph_name or os.path.splitext(
+            os.path.basename(args.part_config))[0]
+    if dataset_dir is None:
+        dataset_dir = os.path.join(args.workspace, args.rel_data_path)
+    assert graph_name, "need --graph-name or --part_config"
     with open(args.hostfile) as f:
         hosts = parse_hostfile(f.read())
-    dispatch_partitions(args.dataset_dir, args.graph_name, hosts,
+    dispatch_partitions(dataset_dir, graph_name, hosts,
                         workspace=args.workspace)
 
 

@@ -265,3 +265,39 @@ def test_ldg_recovers_planted_communities():
     # balance cap held
     sizes = torch.bincount(a, minlength=P)
     assert int(sizes.max()) <= int(n / P * 1.05) + 2
+
+
+def test_dispatch_cli_reference_contract(tmp_path):
+    """dispatch accepts the reference dglrun's invocation spellings
+    (--workspace --rel_data_path --rel_workload_path --part_config
+    --ip_config, reference dglrun:182-188)."""
+    from dgl_operator_amd.tools import dispatch as dispatch_mod
+
+    g = rmat_graph(40, 200, num_feats=2, seed=8)
+    ds = tmp_path / "ws" / "dataset"
+    partition_graph(g, "ref", 2, str(ds), algorithm="range")
+    hostfile = tmp_path / "hostfile"
+    hostfile.write_text(HOSTFILE)
+    import os as _os
+    env_root = str(tmp_path / "pods")
+    old = _os.environ.get("DGL_LOCAL_FABRIC_ROOT")
+    _os.environ["DGL_LOCAL_FABRIC_ROOT"] = env_root
+    try:
+        dispatch_mod.main([
+            "--workspace", str(tmp_path / "ws"),
+            "--rel_data_path", "dataset",
+            "--rel_workload_path", "workload",
+            "--part_config", str(ds / "ref.json"),
+            "--ip_config", str(hostfile),
+        ])
+    finally:
+        if old is None:
+            _os.environ.pop("DGL_LOCAL_FABRIC_ROOT", None)
+        else:
+            _os.environ["DGL_LOCAL_FABRIC_ROOT"] = old
+    for i in range(2):
+        pod = f"job-worker-{i}"
+        assert os.path.exists(
+            os.path.join(env_root, pod,
+                         str(tmp_path / "ws").lstrip("/"),
+                         "workload", f"part{i}", "graph.pt"))
