@@ -182,10 +182,12 @@ def run_launcher(args):
                                 workspace=args.workspace)
     with phase("Phase 4/5 revise hostfile"):
         if args.revise_hostfile_entry_point:
+            # reference invocation contract (dglrun:205)
             launch_mod.exec_batch(
                 hosts,
                 f"python {args.revise_hostfile_entry_point} "
-                f"--hostfile {args.hostfile} --workspace {args.workspace}",
+                f"--workspace {args.workspace} "
+                f"--ip_config {args.hostfile} --framework DGL",
             )
         else:
             revised = revise_for_dgl(hosts)

@@ -47,11 +47,18 @@ def revise_for_dglke(entries: List[HostEntry], num_servers: int = 1) -> str:
 def main(argv=None):
     import argparse
 
+    # accepts BOTH this repo's spellings and the reference
+    # revise_hostfile.py contract (--workspace --ip_config --framework
+    # DGL|DGLKE --num_servers; reference dglrun:205 / dglkerun:258-260)
     p = argparse.ArgumentParser(description="revise hostfile into ipconfig")
-    p.add_argument("--hostfile", default="/etc/dgl/hostfile")
+    p.add_argument("--hostfile", "--ip_config", "--ip-config",
+                   default="/etc/dgl/hostfile")
     p.add_argument("--output", default=None)
-    p.add_argument("--format", choices=["dgl", "dglke"], default="dgl")
-    p.add_argument("--num-servers", type=int, default=1)
+    p.add_argument("--format", "--framework",
+                   type=lambda s: s.lower(),
+                   choices=["dgl", "dglke"], default="dgl")
+    p.add_argument("--num-servers", "--num_servers", type=int, default=1)
+    p.add_argument("--workspace", default=None)
     args = p.parse_args(argv)
     with open(args.hostfile) as f:
         entries = parse_hostfile(f.read())
@@ -61,7 +68,8 @@ def main(argv=None):
         else revise_for_dglke(entries, args.num_servers)
     )
     out = args.output or os.path.join(
-        os.environ.get("WORKSPACE", "."), "hostfile_revised"
+        args.workspace or os.environ.get("WORKSPACE", "."),
+        "hostfile_revised"
     )
     with open(out, "w") as f:
         f.write(text)

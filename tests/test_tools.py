@@ -301,3 +301,22 @@ def test_dispatch_cli_reference_contract(tmp_path):
             os.path.join(env_root, pod,
                          str(tmp_path / "ws").lstrip("/"),
                          "workload", f"part{i}", "graph.pt"))
+
+
+def test_hostfile_cli_reference_contract(tmp_path):
+    """revise_hostfile accepts the reference spellings (--workspace
+    --ip_config --framework DGLKE --num_servers)."""
+    from dgl_operator_amd.tools import hostfile as hf
+
+    src = tmp_path / "hostfile"
+    src.write_text(HOSTFILE)
+    ws = tmp_path / "ws"
+    ws.mkdir()
+    hf.main(["--workspace", str(ws), "--ip_config", str(src),
+             "--framework", "DGLKE", "--num_servers", "3"])
+    out = (ws / "hostfile_revised").read_text()
+    assert out == "10.244.0.5 30050 3\n10.244.0.6 30050 3\n"
+    hf.main(["--workspace", str(ws), "--ip_config", str(src),
+             "--framework", "DGL"])
+    assert (ws / "hostfile_revised").read_text() == \
+        "10.244.0.5 30050\n10.244.0.6 30050\n"
