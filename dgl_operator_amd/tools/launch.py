@@ -78,27 +78,56 @@ def train(
 def main(argv=None):
     import argparse
 
+    # accepts BOTH this repo's spellings and the reference launch.py
+    # invocation contract (python/dglrun/tools/launch.py:158-188 +
+    # the call sites in exec/dglrun:202-234 and exec/dglkerun:190-233):
+    # the command is POSITIONAL there, hosts come from --ip_config, copies
+    # use --source_file_paths/--target_dir (+ --container), and train
+    # carries --num_trainers/--num_samplers/--num_servers/--num_parts.
     p = argparse.ArgumentParser()
     p.add_argument("--cmd_type", required=True,
                    choices=["exec_batch", "copy_batch", "copy_batch_container",
                             "train"])
-    p.add_argument("--hostfile", default="/etc/dgl/hostfile")
+    p.add_argument("--hostfile", "--ip_config", "--ip-config",
+                   default="/etc/dgl/hostfile")
     p.add_argument("--command", default="")
-    p.add_argument("--source", default="")
-    p.add_argument("--target", default="")
+    p.add_argument("--source", "--source_file_paths", default="")
+    p.add_argument("--target", "--target_dir", default="")
     p.add_argument("--container", default=None)
     p.add_argument("--master-port", type=int, default=29400)
     p.add_argument("--script", default="")
     p.add_argument("--script-args", default="")
+    p.add_argument("--workspace", default=None)  # accepted (reference)
+    p.add_argument("--num_trainers", "--num-trainers", type=int, default=None)
+    p.add_argument("--num_samplers", "--num-samplers", type=int, default=0)
+    p.add_argument("--num_servers", "--num-servers", type=int, default=0)
+    p.add_argument("--num_parts", "--num-parts", type=int, default=None)
+    p.add_argument("--part_config", "--part-config", default=None)
+    p.add_argument("positional_command", nargs="?", default="",
+                   help="reference style: the command/script as the last "
+                        "positional argument")
     args = p.parse_args(argv)
     with open(args.hostfile) as f:
         hosts = parse_hostfile(f.read())
+    command = args.command or args.positional_command
     if args.cmd_type == "exec_batch":
-        exec_batch(hosts, args.command)
+        exec_batch(hosts, command)
     elif args.cmd_type in ("copy_batch", "copy_batch_container"):
-        copy_batch(hosts, args.source, args.target, container=args.container)
+        copy_batch(hosts, args.source or command, args.target,
+                   container=args.container)
     elif args.cmd_type == "train":
-        train(hosts, args.script, args.script_args, args.master_port)
+        # reference style passes the whole train command positionally;
+        # repo style splits --script/--script-args
+        script, script_args = args.script, args.script_args
+        if not script and command:
+            script, script_args = command, ""
+        if args.num_parts is not None:
+            total = (args.num_trainers or hosts[0].slots) * len(hosts)
+            assert args.num_parts == total, (
+                f"one partition per rank: --num_parts {args.num_parts} vs "
+                f"{total} ranks")
+        train(hosts, script, script_args, args.master_port,
+              num_trainers=args.num_trainers)
 
 
 if __name__ == "__main__":

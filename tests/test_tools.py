@@ -320,3 +320,31 @@ def test_hostfile_cli_reference_contract(tmp_path):
              "--framework", "DGL"])
     assert (ws / "hostfile_revised").read_text() == \
         "10.244.0.5 30050\n10.244.0.6 30050\n"
+
+
+def test_launch_cli_reference_contract(tmp_path, monkeypatch):
+    """launch.py accepts the reference invocation style: positional
+    command, --ip_config, --source_file_paths/--target_dir
+    (exec/dglrun:202-234, exec/dglkerun:190-233)."""
+    from dgl_operator_amd.tools import launch as launch_mod
+
+    hostfile = tmp_path / "hostfile"
+    hostfile.write_text(HOSTFILE)
+    monkeypatch.setenv("DGL_LOCAL_FABRIC_ROOT", str(tmp_path / "pods"))
+    # exec_batch with positional command (touch a file in each pod dir)
+    launch_mod.main([
+        "--ip_config", str(hostfile), "--cmd_type", "exec_batch",
+        "touch proof.txt",
+    ])
+    for pod in ("job-worker-0", "job-worker-1"):
+        assert (tmp_path / "pods" / pod / "proof.txt").exists()
+    # copy_batch with --source_file_paths/--target_dir
+    src = tmp_path / "payload.txt"
+    src.write_text("data")
+    launch_mod.main([
+        "--ip_config", str(hostfile), "--cmd_type", "copy_batch",
+        "--source_file_paths", str(src), "--target_dir", "/w/payload.txt",
+    ])
+    for pod in ("job-worker-0", "job-worker-1"):
+        assert (tmp_path / "pods" / pod / "w" /
+                "payload.txt").read_text() == "data"
