@@ -28,6 +28,8 @@ JOB_NAME_LABEL = "dgl-job-name"
 ENV_PHASE = "DGL_OPERATOR_PHASE_ENV"
 ENV_KUBEXEC_PATH = "DGL_OPERATOR_KUBEXEC_PATH"
 ENV_KUBECTL_PATH = "DGL_OPERATOR_KUBECTL_PATH"
+ENV_HOSTFILE_PATH = "DGL_OPERATOR_HOSTFILE_PATH"
+ENV_PARTFILE_PATH = "DGL_OPERATOR_PARTFILE_PATH"
 ENV_OPERATOR = "DGL_OPERATOR_ENV"
 CONFIG_MOUNT = "/etc/dgl"
 KUBECTL_MOUNT = "/opt/kube"

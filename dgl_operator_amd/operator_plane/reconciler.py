@@ -27,6 +27,7 @@ from typing import Dict, List
 from .api import (
     CONFIG_MOUNT,
     DGL_PORT,
+    ENV_HOSTFILE_PATH,
     ENV_KUBECTL_PATH,
     ENV_KUBEXEC_PATH,
     ENV_OPERATOR,
@@ -280,6 +281,7 @@ class DGLJobReconciler:
                             {"limits": {"cpu": "1", "memory": "2Gi"}})
         env = dict(pod_spec.get("env", {}))
         env[ENV_KUBEXEC_PATH] = f"{CONFIG_MOUNT}/kubexec.sh"
+        env[ENV_HOSTFILE_PATH] = f"{CONFIG_MOUNT}/hostfile"
         env[ENV_KUBECTL_PATH] = f"{KUBECTL_MOUNT}/kubectl"
         env[ENV_OPERATOR] = "1"
         if job.spec.partition_mode == PartitionMode.SKIP:
@@ -324,24 +326,39 @@ class DGLJobReconciler:
             env[ENV_PHASE] = "Partitioner"
             env[ENV_KUBEXEC_PATH] = f"{CONFIG_MOUNT}/kubexec.sh"
             env[ENV_KUBECTL_PATH] = f"{KUBECTL_MOUNT}/kubectl"
-            env[ENV_OPERATOR] = "1"
         else:
             if "command" not in pod_spec:
                 pod_spec["command"] = ["sleep", "365d"]
+        # every worker-like pod carries DGL_OPERATOR_ENV=1
+        # (dgljob_controller.go:935-939)
+        env[ENV_OPERATOR] = "1"
+        is_part = rtype == ReplicaType.PARTITIONER
+        spec = {
+            **pod_spec,
+            "env": env,
+            # the partitioner has NO ports (go:1026 clears them); it DOES
+            # get the kubectl-download init container + kube volume so its
+            # phase-2 `kubectl cp` delivery works (go:1009-1051)
+            "ports": ([] if is_part
+                      else list(range(DGL_PORT, DGL_PORT + HOST_PORT_NUM))),
+            "volumes": (["config", "shm", "kube"] if is_part
+                        else ["config", "shm"]),
+            # /dev/shm emptyDir sized to half the memory limit
+            # (dgljob_controller.go:961-974)
+            "shmSizeFraction": 0.5,
+        }
+        if is_part:
+            spec["initContainers"] = [{
+                "name": "kubectl-download",
+                "image": self.kubectl_download_image,
+            }]
+            spec["serviceAccount"] = f"{job.name}-partitioner"
         return Pod(
             name=name,
             namespace=job.namespace,
             labels=self._base_labels(job, rtype, name),
             annotations={REPLICA_INDEX_ANNOTATION: str(index)},
-            spec={
-                **pod_spec,
-                "env": env,
-                "ports": list(range(DGL_PORT, DGL_PORT + HOST_PORT_NUM)),
-                "volumes": ["config", "shm"],
-                # /dev/shm emptyDir sized to half the memory limit
-                # (dgljob_controller.go:961-974)
-                "shmSizeFraction": 0.5,
-            },
+            spec=spec,
             owner=job.name,
             owner_uid=job.uid or None,
         )

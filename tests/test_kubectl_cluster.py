@@ -151,6 +151,22 @@ def test_full_ladder_through_kubectl(kube):
     pp = read_obj(state, "pod", ns, f"{name}-partitioner")
     assert lp and pp
     assert lp["spec"]["restartPolicy"] == "Never"
+    # launcher env contract (go:1196-1214)
+    lenv = {e["name"]: e["value"] for e in lp["spec"]["containers"][0]["env"]}
+    assert lenv["DGL_OPERATOR_KUBEXEC_PATH"] == "/etc/dgl/kubexec.sh"
+    assert lenv["DGL_OPERATOR_HOSTFILE_PATH"] == "/etc/dgl/hostfile"
+    assert lenv["DGL_OPERATOR_KUBECTL_PATH"] == "/opt/kube/kubectl"
+    assert lenv["DGL_OPERATOR_ENV"] == "1"
+    # partitioner: kubectl-download init + kube volume, NO ports
+    # (go:1009-1051, :1026)
+    pinits = [ic["name"] for ic in pp["spec"].get("initContainers", [])]
+    assert pinits == ["kubectl-download"]
+    pvols = {v["name"] for v in pp["spec"]["volumes"]}
+    assert "kubectl-volume" in pvols
+    assert not pp["spec"]["containers"][0].get("ports")
+    penv = {e["name"]: e["value"] for e in pp["spec"]["containers"][0]["env"]}
+    assert penv["DGL_OPERATOR_PHASE_ENV"] == "Partitioner"
+    assert penv["DGL_OPERATOR_ENV"] == "1"
     inits = [ic["name"] for ic in lp["spec"]["initContainers"]]
     assert inits == ["kubectl-download", "watcher-loop-partitioner",
                      "watcher-loop-worker"]
@@ -197,6 +213,9 @@ def test_full_ladder_through_kubectl(kube):
             "command" in wp["spec"]["containers"][0]
         ports = wp["spec"]["containers"][0]["ports"]
         assert len(ports) == 20 and ports[0]["containerPort"] == 30050
+        wenv = {e["name"]: e["value"]
+                for e in wp["spec"]["containers"][0]["env"]}
+        assert wenv["DGL_OPERATOR_ENV"] == "1"  # go:935-939
         wvols = {v["name"]: v for v in wp["spec"]["volumes"]}
         assert wvols["dshm"]["emptyDir"]["medium"] == "Memory"
         svc = read_obj(state, "service", ns, f"{name}-worker-{i}")
