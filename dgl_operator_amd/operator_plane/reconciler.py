@@ -149,10 +149,14 @@ class DGLJobReconciler:
     def _ensure_configmap(self, job: DGLJob):
         cm = self.cluster.get_configmap(job.namespace, self.cm_name(job))
         if cm is None:
+            # all four keys exist from creation: the config volume's items
+            # reference them, and a kubelet fails the mount for a missing
+            # key (they fill in as pod IPs appear)
             cm = ConfigMap(
                 name=self.cm_name(job),
                 namespace=job.namespace,
-                data={"kubexec.sh": KUBEXEC_SH},
+                data={"kubexec.sh": KUBEXEC_SH, "hostfile": "",
+                      "partfile": "", "leadfile": ""},
                 owner=job.name,
                 owner_uid=job.uid or None,
             )
@@ -172,16 +176,21 @@ class DGLJobReconciler:
                 host_lines.append(
                     f"{p.ip} {DGL_PORT} {p.name} slots={job.spec.slots_per_worker}"
                 )
-        cm.data["hostfile"] = "\n".join(host_lines) + ("\n" if host_lines else "")
+        new = dict(cm.data)
+        new["hostfile"] = "\n".join(host_lines) + ("\n" if host_lines else "")
+        # partfile/leadfile are the reference's 3-column format (no slots=,
+        # dgljob_controller.go:1440-1469)
         part = pods.get(job.partitioner_name())
-        cm.data["partfile"] = (
-            f"{part.ip} {DGL_PORT} {part.name} slots=1\n" if part and part.ip else ""
+        new["partfile"] = (
+            f"{part.ip} {DGL_PORT} {part.name}\n" if part and part.ip else ""
         )
         lead = pods.get(job.launcher_name())
-        cm.data["leadfile"] = (
-            f"{lead.ip} {DGL_PORT} {lead.name} slots=1\n" if lead and lead.ip else ""
+        new["leadfile"] = (
+            f"{lead.ip} {DGL_PORT} {lead.name}\n" if lead and lead.ip else ""
         )
-        self.cluster.update_configmap(cm)
+        if new != cm.data:  # update in place only on change (go:1431-1436)
+            cm.data = new
+            self.cluster.update_configmap(cm)
 
     # -- RBAC -----------------------------------------------------------
     def _ensure_rbac(self, job: DGLJob, needs_partitioner: bool):

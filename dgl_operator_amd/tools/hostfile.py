@@ -20,18 +20,20 @@ class HostEntry:
 
 
 def parse_hostfile(text: str) -> List[HostEntry]:
+    """Hostfile lines are `ip port podname slots=N`; partfile/leadfile
+    lines are the reference's 3-column `ip port podname`
+    (dgljob_controller.go:1440-1469) — slots defaults to 1 there."""
     out = []
     for line in text.splitlines():
         parts = line.split()
-        if len(parts) < 4:
+        if len(parts) < 3:
             continue
+        slots = 1
+        if len(parts) >= 4 and "=" in parts[3]:
+            slots = int(parts[3].split("=")[1])
         out.append(
-            HostEntry(
-                ip=parts[0],
-                port=int(parts[1]),
-                pod=parts[2],
-                slots=int(parts[3].split("=")[1]),
-            )
+            HostEntry(ip=parts[0], port=int(parts[1]), pod=parts[2],
+                      slots=slots)
         )
     return out
 
