@@ -34,6 +34,9 @@ __version__ = "0.2.0"
 try:  # compute plane needs torch; the control-plane-only manager image
     from . import graph  # noqa: F401
     from . import fn  # noqa: F401
-    from .graph import Graph, Block, batch_graphs  # noqa: F401
+    from .graph import (  # noqa: F401
+        Graph, Block, NID, EID, batch_graphs, batch_num_edges,
+        to_bidirected, unbatch,
+    )
 except ImportError:  # pragma: no cover
     graph = None

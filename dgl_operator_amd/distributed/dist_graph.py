@@ -304,6 +304,18 @@ class DistGraph:
     def barrier(self):
         comm.barrier()
 
+    # -- DGL-name conveniences (reference train_dist.py uses g.rank(),
+    # g.get_partition_book(), g.number_of_nodes()) -----------------------
+    def get_partition_book(self):
+        return self.book
+
+    def number_of_nodes(self) -> int:
+        return self.book.num_nodes
+
+    @property
+    def num_nodes(self) -> int:
+        return self.book.num_nodes
+
 
 # ---------------------------------------------------------------------------
 # Ghost-zone (halo) replication — the MI355X answer to the reference's

@@ -1,4 +1,13 @@
-from .graph import Graph, Block, batch_graphs
+from .graph import (
+    EID,
+    NID,
+    Block,
+    Graph,
+    batch_graphs,
+    batch_num_edges,
+    to_bidirected,
+    unbatch,
+)
 from .rmat import rmat_edges, rmat_graph, ogbn_products_shape
 from .partition import (
     partition_graph,
@@ -9,7 +18,12 @@ from .partition import (
 __all__ = [
     "Graph",
     "Block",
+    "NID",
+    "EID",
     "batch_graphs",
+    "batch_num_edges",
+    "to_bidirected",
+    "unbatch",
     "rmat_edges",
     "rmat_graph",
     "ogbn_products_shape",

@@ -35,6 +35,12 @@ def _coo_to_compressed(
     return indptr, cols[perm], perm
 
 
+# DGL's special field names (dgl.NID / dgl.EID are both "_ID"): blocks
+# carry their original node ids under srcdata[NID]/dstdata[NID]
+NID = "_ID"
+EID = "_ID"
+
+
 class Graph:
     """Homogeneous directed graph. Edges point src -> dst; messages flow along edges."""
 
@@ -126,6 +132,43 @@ class Graph:
             g._csr = tuple(t.to(device) for t in self._csr)
         g.ndata = {k: v.to(device) for k, v in self.ndata.items()}
         g.edata = {k: v.to(device) for k, v in self.edata.items()}
+        return g
+
+    # -- DGL-name conveniences (dgl.DGLGraph.number_of_* etc.) -------------
+    def number_of_nodes(self) -> int:
+        return self._num_nodes
+
+    def number_of_edges(self) -> int:
+        return self.num_edges
+
+    def local_scope(self):
+        """Context manager isolating ndata/edata ASSIGNMENTS (DGL
+        g.local_scope(), used by the reference conv tutorials): dict
+        mutations inside the block are rolled back on exit. In-place
+        tensor mutation is not isolated — same caveat as DGL."""
+        import contextlib
+
+        @contextlib.contextmanager
+        def scope():
+            nd, ed = dict(self.ndata), dict(self.edata)
+            try:
+                yield self
+            finally:
+                self.ndata = nd
+                self.edata = ed
+
+        return scope()
+
+    def remove_edges(self, eids: torch.Tensor) -> "Graph":
+        """New graph without the given edge positions (dgl.remove_edges —
+        the reference link-predict tutorial removes test edges); ndata is
+        carried over, edata rows are filtered."""
+        keep = torch.ones(self.num_edges, dtype=torch.bool,
+                          device=self._src.device)
+        keep[eids] = False
+        g = Graph(self._src[keep], self._dst[keep], self._num_nodes)
+        g.ndata = dict(self.ndata)
+        g.edata = {k: v[keep] for k, v in self.edata.items()}
         return g
 
     def add_self_loops(self) -> "Graph":
@@ -256,6 +299,11 @@ class Block:
         self.edata = {}
         self.srcdata = {}  # per-src-node fields (block-local rows)
         self.dstdata = {}  # per-dst-node fields
+        if srcdata_nids is not None:
+            # DGL convention: block.srcdata[dgl.NID] / dstdata[dgl.NID]
+            # are the original node ids (dst nodes are the first num_dst)
+            self.srcdata[NID] = srcdata_nids
+            self.dstdata[NID] = srcdata_nids[:num_dst]
 
     @property
     def num_src_nodes(self) -> int:
@@ -347,3 +395,49 @@ def batch_graphs(graphs: Sequence[Graph]) -> Tuple[Graph, torch.Tensor]:
         off += g.num_nodes
     bg = Graph(torch.cat(srcs), torch.cat(dsts), off)
     return bg, torch.tensor(sizes, dtype=torch.int64)
+
+
+def unbatch(bg: Graph, batch_num_nodes: torch.Tensor) -> List[Graph]:
+    """Split a batched graph back into components (dgl.unbatch — the
+    graph-classification tutorial's inverse of dgl.batch). Edges belong to
+    the component owning their endpoints (disjoint union => src comp ==
+    dst comp)."""
+    bounds = torch.zeros(batch_num_nodes.numel() + 1, dtype=torch.int64)
+    torch.cumsum(batch_num_nodes, 0, out=bounds[1:])
+    src, dst = bg.edges()
+    comp = torch.bucketize(src, bounds[1:-1], right=True)
+    out = []
+    for i in range(batch_num_nodes.numel()):
+        m = comp == i
+        lo = int(bounds[i])
+        g = Graph(src[m] - lo, dst[m] - lo, int(batch_num_nodes[i]))
+        g.ndata = {k: v[lo : int(bounds[i + 1])]
+                   for k, v in bg.ndata.items()}
+        g.edata = {k: v[m] for k, v in bg.edata.items()}
+        out.append(g)
+    return out
+
+
+def batch_num_edges(bg: Graph, batch_num_nodes: torch.Tensor) -> torch.Tensor:
+    """Edges per component of a batched graph (dgl.batch_num_edges)."""
+    bounds = torch.zeros(batch_num_nodes.numel() + 1, dtype=torch.int64)
+    torch.cumsum(batch_num_nodes, 0, out=bounds[1:])
+    src, _ = bg.edges()
+    comp = torch.bucketize(src, bounds[1:-1], right=True)
+    return torch.bincount(comp, minlength=batch_num_nodes.numel())
+
+
+def to_bidirected(g: Graph, copy_ndata: bool = True) -> Graph:
+    """Union of g's edges and their reverses, deduplicated
+    (dgl.to_bidirected, used by the reference link-predict tutorial)."""
+    src, dst = g.edges()
+    s = torch.cat([src, dst])
+    d = torch.cat([dst, src])
+    key = s * g.num_nodes + d
+    uniq, first = torch.unique(key, return_inverse=False), None
+    us = uniq // g.num_nodes
+    ud = uniq % g.num_nodes
+    out = Graph(us, ud, g.num_nodes)
+    if copy_ndata:
+        out.ndata = dict(g.ndata)
+    return out

@@ -100,3 +100,54 @@ def test_block_device_roundtrip():
     assert torch.equal(b2.srcdata_nids, b.srcdata_nids)
     assert torch.equal(b2.edata["w"], b.edata["w"])
     assert b2.num_src_nodes == 3 and b2.num_dst_nodes == 2
+
+
+def test_dgl_api_conveniences():
+    """DGL-name surface the reference tutorials use: number_of_*,
+    local_scope, remove_edges, to_bidirected, unbatch, NID on blocks."""
+    import dgl_operator_amd as doa
+    from dgl_operator_amd.graph import (
+        batch_graphs, batch_num_edges, to_bidirected, unbatch,
+    )
+
+    g = Graph(torch.tensor([0, 1, 2]), torch.tensor([1, 2, 0]), 4)
+    assert g.number_of_nodes() == 4 and g.number_of_edges() == 3
+    # local_scope rolls back assignments
+    g.ndata["h"] = torch.ones(4)
+    with g.local_scope():
+        g.ndata["tmp"] = torch.zeros(4)
+        g.edata["w"] = torch.ones(3)
+    assert "tmp" not in g.ndata and "w" not in g.edata
+    assert "h" in g.ndata
+    # remove_edges drops positions + edata rows
+    g.edata["w"] = torch.arange(3.0)
+    g2 = g.remove_edges(torch.tensor([1]))
+    assert g2.num_edges == 2
+    assert torch.equal(g2.edata["w"], torch.tensor([0.0, 2.0]))
+    # to_bidirected dedups reverses
+    gb = to_bidirected(g)
+    assert gb.num_edges == 6
+    gb2 = to_bidirected(gb)
+    assert gb2.num_edges == 6
+    # batch/unbatch round trip
+    gs = [Graph(torch.tensor([0]), torch.tensor([1]), 2),
+          Graph(torch.tensor([0, 1]), torch.tensor([1, 2]), 3)]
+    gs[0].ndata["x"] = torch.tensor([1.0, 2.0])
+    gs[1].ndata["x"] = torch.tensor([3.0, 4.0, 5.0])
+    bg, bnn = batch_graphs(gs)
+    bg.ndata["x"] = torch.cat([gs[0].ndata["x"], gs[1].ndata["x"]])
+    assert torch.equal(batch_num_edges(bg, bnn), torch.tensor([1, 2]))
+    parts = unbatch(bg, bnn)
+    assert len(parts) == 2
+    assert parts[1].num_nodes == 3 and parts[1].num_edges == 2
+    assert torch.equal(parts[1].ndata["x"], gs[1].ndata["x"])
+    s, d = parts[1].edges()
+    assert torch.equal(s, torch.tensor([0, 1]))
+    # Block NID fields
+    from dgl_operator_amd.graph import Block, NID
+
+    blk = Block(torch.tensor([0, 1, 2]), torch.tensor([1, 2]), 3, 2,
+                srcdata_nids=torch.tensor([10, 20, 30]))
+    assert torch.equal(blk.srcdata[NID], torch.tensor([10, 20, 30]))
+    assert torch.equal(blk.dstdata[NID], torch.tensor([10, 20]))
+    assert doa.NID == "_ID"
