@@ -429,3 +429,47 @@ def test_manifest_roundtrip_serialization():
         for rt, rs in job.spec.replica_specs.items():
             assert job2.spec.replica_specs[rt].replicas == rs.replicas
             assert job2.spec.replica_specs[rt].template == rs.template
+
+
+def test_reconciler_random_soak():
+    """Randomized kubelet-event soak: arbitrary phase flips, deletions and
+    evictions interleaved with reconciles must never crash the reconciler,
+    always produce well-formed statuses, and terminal cleanup must honor
+    cleanPodPolicy=All."""
+    import random
+
+    rng = random.Random(7)
+    for trial in range(40):
+        mode = rng.choice(["DGL-API", "Skip", "ParMETIS"])
+        workers = rng.randint(1, 4)
+        yaml_doc = GRAPHSAGE_YAML.replace("DGL-API", mode).replace(
+            "replicas: 2", f"replicas: {workers}")
+        job = job_from_manifest(yaml_doc)
+        job.spec.clean_pod_policy = CleanPodPolicy.ALL
+        c = FakeCluster()
+        r = DGLJobReconciler(c)
+        phases = ["Pending", "Running", "Succeeded", "Failed"]
+        for step in range(rng.randint(3, 15)):
+            r.reconcile(job)
+            pods = list(c.pods.values())
+            if pods and rng.random() < 0.8:
+                p = rng.choice(pods)
+                ph = PodPhase(rng.choice(phases))
+                reason = "Evicted" if (ph == PodPhase.FAILED
+                                       and rng.random() < 0.3) else None
+                c.set_pod_phase(p.namespace, p.name, ph, reason=reason)
+            if pods and rng.random() < 0.1:
+                p = rng.choice(pods)
+                c.delete_pod(p.namespace, p.name)
+            # status always well-formed
+            for st in job.status.replica_statuses.values():
+                got, total = st.ready.split("/")
+                assert 0 <= int(got) <= int(total)
+                assert st.active >= 0 and st.failed >= 0
+            assert job.status.phase is None or isinstance(
+                job.status.phase, JobPhase)
+        # CR deletion: cleanup must remove every pod under policy All
+        job.deletion_timestamp = 1.0
+        r.reconcile(job)
+        assert c.list_pods(job.namespace, job.name) == [], (
+            trial, mode, workers)
