@@ -330,3 +330,25 @@ def test_leader_election(kube):
     # a contends again but the lease is freshly held by b
     a.reconcile_once()
     assert not a.is_leader
+
+
+def test_shm_sizing_and_quantities():
+    """/dev/shm emptyDir = half the container memory limit
+    (dgljob_controller.go:961-974), via the quantity parser."""
+    from dgl_operator_amd.operator_plane.k8s import parse_quantity, pod_manifest
+
+    assert parse_quantity("20Gi") == 20 * 2**30
+    assert parse_quantity("2G") == 2 * 10**9
+    assert parse_quantity("512Mi") == 512 * 2**20
+    assert parse_quantity(123) == 123
+    pod = Pod("w", "ns", spec={
+        "containers": [{"name": "c", "image": "i",
+                        "resources": {"limits": {"memory": "20Gi"}}}],
+        "volumes": ["config", "shm"],
+        "shmSizeFraction": 0.5,
+    }, owner="j")
+    m = pod_manifest(pod)
+    dshm = [v for v in m["spec"]["volumes"] if v["name"] == "dshm"][0]
+    # 10Gi ≈ 10.7 G (decimal) -> floor to 10G
+    assert dshm["emptyDir"]["sizeLimit"] == "10G"
+    assert dshm["emptyDir"]["medium"] == "Memory"
