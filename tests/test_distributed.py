@@ -397,3 +397,33 @@ def _loader_worker(rank, world):
 
 def test_dist_node_dataloader():
     _run_workers(_loader_worker)
+
+
+def test_reorder_segments_unit():
+    """Direct contract test of the alltoallv segment reorder helper
+    (empty segments, non-trivial permutation, variable lengths)."""
+    from dgl_operator_amd.distributed.dist_graph import (
+        _reorder_segments, _segment_sum_by_rank,
+    )
+
+    # original segments: [ [10,11], [], [20], [30,31,32] ]
+    # sorted order visits them as perm = positions of sorted in original
+    counts_orig = torch.tensor([2, 0, 1, 3])
+    perm = torch.tensor([3, 0, 2, 1])  # sorted seg s belongs at perm[s]
+    counts_sorted = counts_orig[perm]
+    segs = {0: [10, 11], 1: [], 2: [20], 3: [30, 31, 32]}
+    payload_sorted = torch.tensor(
+        [v for s in perm.tolist() for v in segs[s]])
+    out, counts_back = _reorder_segments(payload_sorted, counts_sorted, perm)
+    assert torch.equal(counts_back, counts_orig)
+    assert torch.equal(out, torch.tensor([10, 11, 20, 30, 31, 32]))
+    # all-empty payload path
+    out2, c2 = _reorder_segments(torch.empty(0, dtype=torch.int64),
+                                 torch.zeros(3, dtype=torch.int64),
+                                 torch.tensor([2, 0, 1]))
+    assert out2.numel() == 0 and torch.equal(c2, torch.zeros(3).long())
+    # segment sums by rank
+    v = torch.tensor([1.0, 2.0, 3.0, 4.0])
+    assert torch.equal(
+        _segment_sum_by_rank(v, torch.tensor([1, 0, 3])),
+        torch.tensor([1.0, 0.0, 9.0]))
