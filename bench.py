@@ -491,6 +491,12 @@ def main():
                 "step_mode": ("hipGraph-captured" if use_capture
                               else ("eager+prefetch" if use_prefetch
                                     else "eager")),
+                # BASELINE metric's companion number: one epoch = every
+                # owned node seeded once (ceil(nodes/ws/batch) steps/rank)
+                "epoch_time_s": round(
+                    (elapsed / args.steps)
+                    * ((args.nodes // ws + args.batch - 1) // args.batch),
+                    3),
             },
         }))
 
