@@ -51,9 +51,10 @@ def parse_args():
     p.add_argument("--lr", type=float, default=3e-3)
     p.add_argument("--dropout", type=float, default=0.5)
     p.add_argument("--partition", type=str, default="range",
-                   choices=["range"],
-                   help="in-bench sharding is contiguous ranges; LDG layouts "
-                        "come from the partitioner pod flow (graph/partition.py)")
+                   choices=["range", "ldg"],
+                   help="in-bench sharding: contiguous ranges (default) or "
+                        "the LDG streaming partitioner (smaller halos at "
+                        "ws>1; adds a CPU partitioning pass at setup)")
     p.add_argument("--no-halo", action="store_true",
                    help="disable ghost-zone replication (fall back to "
                         "per-step alltoallv sampling + feature pulls)")
@@ -241,7 +242,26 @@ def main():
         seed=0, device=device,
     )
     n = g.num_nodes
-    bounds = [n * p // ws for p in range(ws + 1)]
+    if args.partition == "ldg" and ws > 1:
+        # deterministic on every rank: same graph -> same assignment ->
+        # same relabel; shards then follow LDG ownership (smaller halos)
+        from dgl_operator_amd.graph.partition import (
+            ldg_assignment, relabel_by_assignment,
+        )
+
+        t0 = time.time()
+        g_cpu = g.to("cpu") if device.type != "cpu" else g
+        assign = ldg_assignment(g_cpu, ws)
+        g2, bounds, _ = relabel_by_assignment(g_cpu, assign, num_parts=ws)
+        g = g2.to(device) if device.type != "cpu" else g2
+        if rank == 0:
+            s_, d_ = g_cpu.edges()
+            cut = float((assign[s_] != assign[d_]).float().mean())
+            print(f"# ldg partition: {time.time() - t0:.1f}s, "
+                  f"edge-cut {cut:.3f}")
+        del g_cpu, g2
+    else:
+        bounds = [n * p // ws for p in range(ws + 1)]
     book = PartitionBook(bounds, device=device)
     dg = DistGraph.from_full_graph(g, book, rank)
     # free the full graph copies we no longer need (features stay sharded)

@@ -127,6 +127,36 @@ def _assign_ldg(g: Graph, num_parts: int, balance_train: bool = False,
     return _assign_ldg_python(g, num_parts)
 
 
+def relabel_by_assignment(g: Graph, assign: torch.Tensor,
+                          num_parts: "int | None" = None):
+    """Relabel nodes so each part owns one contiguous id range. Returns
+    (relabeled_graph, boundaries, new_of_old). ndata rows are permuted;
+    the same relabel partition_graph applies before writing parts."""
+    if num_parts is None:
+        num_parts = int(assign.max()) + 1 if assign.numel() else 1
+    perm = torch.argsort(assign, stable=True)  # new id -> old id
+    new_of_old = torch.empty_like(perm)
+    new_of_old[perm] = torch.arange(g.num_nodes, dtype=torch.int64)
+    counts = torch.bincount(assign, minlength=num_parts)
+    boundaries = torch.zeros(num_parts + 1, dtype=torch.int64)
+    boundaries[1:] = torch.cumsum(counts, 0)
+    src, dst = g.edges()
+    dev = src.device
+    new_dev = new_of_old.to(dev)
+    g2 = Graph(new_dev[src], new_dev[dst], g.num_nodes)
+    perm_dev = perm.to(dev)
+    g2.ndata = {k: v[perm_dev] for k, v in g.ndata.items()}
+    g2.edata = dict(g.edata)
+    return g2, boundaries.tolist(), new_of_old
+
+
+def ldg_assignment(g: Graph, num_parts: int, balance_train: bool = False,
+                   balance_edges: bool = True) -> torch.Tensor:
+    """Public LDG node->part assignment (CPU; the bench's --partition ldg
+    path and anything else that wants in-memory sharding)."""
+    return _assign_ldg(g, num_parts, balance_train, balance_edges)
+
+
 def partition_graph(
     g: Graph,
     name: str,

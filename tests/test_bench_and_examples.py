@@ -39,6 +39,26 @@ def test_bench_json_contract():
 
 
 @pytest.mark.timeout(400)
+def test_bench_two_rank_ldg_partition():
+    """--partition ldg at ws=2: deterministic in-bench LDG sharding
+    (relabel + boundaries) through the same torchrun launch shape."""
+    args = [sys.executable, "-m", "torch.distributed.run", "--standalone",
+            "--local-addr", "127.0.0.1", "--nproc-per-node", "2",
+            os.path.join(REPO, "bench.py"), "--gpus", "2",
+            "--steps", "2", "--warmup", "1", "--partition", "ldg",
+            "--nodes", "20000", "--edges", "100000", "--batch", "200"]
+    for attempt in range(2):
+        r = _run(args, timeout=360)
+        if r.returncode == 0:
+            break
+    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+    assert "# ldg partition:" in r.stdout
+    d = json.loads([l for l in r.stdout.splitlines()
+                    if l.startswith("{")][-1])
+    assert d["n_gpus"] == 2 and d["value"] > 0
+
+
+@pytest.mark.timeout(400)
 @pytest.mark.parametrize("halo", [True, False])
 def test_bench_two_rank_driver_contract(halo):
     """The exact launch shape the driver uses for N>1 (torchrun, one rank
