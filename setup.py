@@ -9,7 +9,7 @@ from setuptools import find_packages, setup
 
 setup(
     name="dgl-operator-amd",
-    version="0.1.0",
+    version="0.2.0",
     description=(
         "MI355X-native distributed-GNN framework + DGLJob operator "
         "(HIP/CDNA4 kernels, RCCL over xGMI)"
@@ -21,4 +21,14 @@ setup(
     python_requires=">=3.10",
     install_requires=["torch", "pyyaml", "numpy"],
     extras_require={"operator": ["prometheus_client"]},
+    entry_points={
+        "console_scripts": [
+            # the reference images expose dglrun/dglkerun as executables
+            # (python/dglrun/exec/); same names after pip install
+            "dglrun=dgl_operator_amd.tools.dglrun:main",
+            "dglkerun=dgl_operator_amd.tools.dglkerun:main",
+            "dgl-operator-manager=dgl_operator_amd.operator_plane.manager:main",
+            "dgl-watcher-loop=dgl_operator_amd.operator_plane.watcher:main",
+        ],
+    },
 )
