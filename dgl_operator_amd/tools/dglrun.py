@@ -79,8 +79,8 @@ def build_parser():
                         "(--dataset-url); offline partitioners fall back "
                         "to synthetic data")
     p.add_argument("--launch-entry-point", default="",
-                   help="accepted for reference-CLI parity (built-in "
-                        "launcher is used)")
+                   help="custom launch script for Phase 5, invoked with the "
+                        "reference launch.py contract (default: built-in)")
     p.add_argument("--revise-hostfile-entry-point", default="",
                    help="custom hostfile-revise script (default: built-in)")
     p.add_argument("--workspace", "--worksapce", default=os.environ.get(
@@ -204,14 +204,27 @@ def run_launcher(args):
             targs += f" --num-epochs {args.num_epochs}"
         if args.batch_size is not None:
             targs += f" --batch-size {args.batch_size}"
-        launch_mod.train(
-            hosts,
-            args.train_entry_point,
-            f"--graph-name {args.graph_name} "
-            f"--part-config {part_cfg} {targs}",
-            master_port=args.master_port,
-            num_trainers=args.num_trainers,
-        )
+        if args.launch_entry_point:
+            # custom launcher, reference invocation contract
+            # (exec/dglrun:220-234): train command passed positionally
+            train_cmd = (f"python {args.train_entry_point} "
+                         f"--graph-name {args.graph_name} "
+                         f"--part-config {part_cfg} {targs}")
+            lep_args = (f"--workspace {args.workspace} "
+                        f"--ip_config {args.hostfile} --cmd_type train")
+            if args.num_trainers:
+                lep_args += f" --num_trainers {args.num_trainers}"
+            _run(f"python {args.launch_entry_point} {lep_args} "
+                 f"'{train_cmd}'")
+        else:
+            launch_mod.train(
+                hosts,
+                args.train_entry_point,
+                f"--graph-name {args.graph_name} "
+                f"--part-config {part_cfg} {targs}",
+                master_port=args.master_port,
+                num_trainers=args.num_trainers,
+            )
 
 
 def main(argv=None):
