@@ -9,6 +9,7 @@ one rank per GPU (slots), rendezvous at worker 0.
 """
 from __future__ import annotations
 
+import os
 import threading
 from typing import Dict, List, Optional
 
@@ -116,11 +117,21 @@ def main(argv=None):
         copy_batch(hosts, args.source or command, args.target,
                    container=args.container)
     elif args.cmd_type == "train":
-        # reference style passes the whole train command positionally;
-        # repo style splits --script/--script-args
+        # reference style passes the whole train command positionally
+        # ("DGLBACKEND=pytorch python train.py --flag ..."); the reference
+        # launcher strips the interpreter before wrapping with
+        # torch.distributed.launch (launch.py:135-152) — same here
         script, script_args = args.script, args.script_args
         if not script and command:
-            script, script_args = command, ""
+            import shlex
+
+            toks = shlex.split(command)
+            while toks and "=" in toks[0] and not toks[0].startswith("-"):
+                toks.pop(0)  # leading ENV=val assignments
+            if toks and os.path.basename(toks[0]) in ("python", "python3"):
+                toks.pop(0)
+            assert toks, f"empty train command: {command!r}"
+            script, script_args = toks[0], " ".join(toks[1:])
         if args.num_parts is not None:
             total = (args.num_trainers or hosts[0].slots) * len(hosts)
             assert args.num_parts == total, (

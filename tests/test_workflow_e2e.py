@@ -146,3 +146,50 @@ def test_train_dist_checkpoint_resume(tmp_path):
     assert "resumed from epoch 0" in r2.stdout
     assert "Epoch 001 time" in r2.stdout
     assert "Epoch 000 time" not in r2.stdout  # skipped the finished epoch
+
+
+@pytest.mark.timeout(300)
+def test_dglrun_custom_launch_entry_point(tmp_path):
+    """--launch-entry-point: Phase 5 goes through a user-supplied launcher
+    invoked with the reference contract (positional train command,
+    --ip_config, --cmd_type train)."""
+    dataset = tmp_path / "ws" / "dataset"
+    dataset.mkdir(parents=True)
+    r = subprocess.run(
+        [sys.executable,
+         os.path.join(REPO, "examples/graphsage_dist/load_and_partition_graph.py"),
+         "--graph-name", "toy", "--num-partitions", "2",
+         "--output", str(dataset),
+         "--nodes", "300", "--edges", "2500", "--feat", "8",
+         "--classes", "3", "--algorithm", "range"],
+        capture_output=True, text=True, cwd=REPO)
+    assert r.returncode == 0, r.stderr
+    hostfile = tmp_path / "hostfile"
+    hostfile.write_text("127.0.0.1 30050 job-worker-0 slots=1\n"
+                        "127.0.0.1 30050 job-worker-1 slots=1\n")
+    # the user's custom launcher: a thin wrapper over a launch.py-contract
+    # implementation (stands in for a reference-written script)
+    wrapper = tmp_path / "my_launch.py"
+    wrapper.write_text(
+        "from dgl_operator_amd.tools.launch import main\nmain()\n")
+    env = dict(os.environ)
+    env["DGL_LOCAL_FABRIC_ROOT"] = str(tmp_path / "pods")
+    env.pop("DGL_OPERATOR_PHASE_ENV", None)
+    env["PYTHONPATH"] = REPO
+    for attempt in range(2):
+        r = subprocess.run(
+            [sys.executable, "-m", "dgl_operator_amd.tools.dglrun",
+             "--graph-name", "toy", "--workspace", "ws",
+             "--hostfile", str(hostfile),
+             "--master-port", str(_free_port()),
+             "--launch-entry-point", str(wrapper),
+             "--train-entry-point",
+             os.path.join(REPO, "examples/graphsage_dist/train_dist.py"),
+             "--train-entry-args",
+             "--num-epochs 1 --batch-size 32 --fan-out 3,3 --log-every 100"],
+            capture_output=True, text=True, cwd=str(tmp_path), env=env,
+            timeout=240)
+        if r.returncode == 0:
+            break
+    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+    assert "Epoch 000" in r.stdout
