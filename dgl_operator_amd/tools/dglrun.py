@@ -211,7 +211,8 @@ def run_launcher(args):
                          f"--graph-name {args.graph_name} "
                          f"--part-config {part_cfg} {targs}")
             lep_args = (f"--workspace {args.workspace} "
-                        f"--ip_config {args.hostfile} --cmd_type train")
+                        f"--ip_config {args.hostfile} --cmd_type train "
+                        f"--master-port {args.master_port}")
             if args.num_trainers:
                 lep_args += f" --num_trainers {args.num_trainers}"
             _run(f"python {args.launch_entry_point} {lep_args} "
