@@ -33,7 +33,9 @@ class LocalPodRuntime:
     def __init__(self, cluster: FakeCluster, root: str,
                  extra_env: Optional[Dict[str, str]] = None):
         self.cluster = cluster
-        self.root = root
+        # pods run with their pod dir as cwd and resolve the fabric root
+        # from env — a relative root would resolve against EACH pod's cwd
+        self.root = os.path.abspath(root)
         self.procs: Dict[str, subprocess.Popen] = {}
         self.started: set = set()
         self.extra_env = extra_env or {}
@@ -159,10 +161,18 @@ def run_job(manifest: str, root: str, timeout: float = 600.0,
     runtime = LocalPodRuntime(mgr.cluster, root, extra_env=extra_env)
     job = mgr.submit(manifest)
     deadline = time.time() + timeout
+    last_phase = None
     try:
         while time.time() < deadline:
             mgr.reconcile_once()
             runtime.tick()
+            if job.status.phase != last_phase:
+                last_phase = job.status.phase
+                ready = {rt.value: rs.ready for rt, rs
+                         in job.status.replica_statuses.items()}
+                print(f"[operator] {job.name} phase -> "
+                      f"{last_phase.value if last_phase else None} {ready}",
+                      flush=True)
             if job.status.phase in (JobPhase.COMPLETED, JobPhase.FAILED):
                 return job
             time.sleep(poll)
