@@ -69,8 +69,10 @@ def build_parser():
     p.add_argument("--partitioned-dataset-dir", default="")
     p.add_argument("--workspace", "--worksapce", default=os.environ.get(
         "WORKSPACE", "/dgl_workspace"))
-    p.add_argument("--hostfile", default="/etc/dgl/hostfile")
-    p.add_argument("--leadfile", default="/etc/dgl/leadfile")
+    p.add_argument("--hostfile", default=os.environ.get(
+        "DGL_OPERATOR_HOSTFILE_PATH", "/etc/dgl/hostfile"))
+    p.add_argument("--leadfile", default=os.environ.get(
+        "DGL_OPERATOR_LEADFILE_PATH", "/etc/dgl/leadfile"))
     p.add_argument("--master-port", type=int, default=29401)
     p.add_argument("--num-triples", type=int, default=100_000)
     p.add_argument("--num-entities", type=int, default=10_000)

@@ -85,8 +85,10 @@ def build_parser():
                    help="custom hostfile-revise script (default: built-in)")
     p.add_argument("--workspace", "--worksapce", default=os.environ.get(
         "WORKSPACE", "/dgl_workspace"))
-    p.add_argument("--hostfile", default="/etc/dgl/hostfile")
-    p.add_argument("--leadfile", default="/etc/dgl/leadfile")
+    p.add_argument("--hostfile", default=os.environ.get(
+        "DGL_OPERATOR_HOSTFILE_PATH", "/etc/dgl/hostfile"))
+    p.add_argument("--leadfile", default=os.environ.get(
+        "DGL_OPERATOR_LEADFILE_PATH", "/etc/dgl/leadfile"))
     p.add_argument("--master-port", type=int, default=29400)
     p.add_argument("--ignore-partition", action="store_true",
                    help="reuse an existing partition under workspace/dataset "

@@ -90,7 +90,8 @@ def main(argv=None):
                    choices=["exec_batch", "copy_batch", "copy_batch_container",
                             "train"])
     p.add_argument("--hostfile", "--ip_config", "--ip-config",
-                   default="/etc/dgl/hostfile")
+                   default=os.environ.get("DGL_OPERATOR_HOSTFILE_PATH",
+                                          "/etc/dgl/hostfile"))
     p.add_argument("--command", default="")
     p.add_argument("--source", "--source_file_paths", default="")
     p.add_argument("--target", "--target_dir", default="")
