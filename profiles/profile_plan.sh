@@ -12,16 +12,18 @@ cd "$GRAFT_REPO_ROOT" || cd /root/repo
 OUT=gpurun_out/prof
 mkdir -p "$OUT"
 
-# 0. eager-vs-captured step A/B + eager phase breakdown (no profiler)
-timeout 300 python bench.py --steps 20 --warmup 5 --no-capture --phase-timing \
-    > "$OUT/bench_eager.log" 2>&1
-timeout 300 python bench.py --steps 20 --warmup 5 \
+# 0. step-mode A/B: plain eager vs eager+prefetch vs captured
+timeout 300 python bench.py --steps 50 --warmup 10 --no-capture --no-prefetch \
+    --phase-timing > "$OUT/bench_eager.log" 2>&1
+timeout 300 python bench.py --steps 50 --warmup 10 --no-capture \
+    > "$OUT/bench_prefetch.log" 2>&1
+timeout 300 python bench.py --steps 50 --warmup 10 \
     > "$OUT/bench_capture.log" 2>&1
 
 # 1. kernel-trace + stats over a short bench run (per-kernel wall time)
 rocprofv3 --kernel-trace --stats -d "$OUT/bench_trace" -- \
     timeout 300 python bench.py --steps 10 --warmup 3 --no-capture \
-    > "$OUT/bench_trace.log" 2>&1
+    --no-prefetch > "$OUT/bench_trace.log" 2>&1
 
 # 2. PMC counters for the SpMM/sampler kernels (separate run, counters only)
 rocprofv3 --pmc SQ_WAVES,SQ_BUSY_CYCLES,TCC_HIT_sum,TCC_MISS_sum \
