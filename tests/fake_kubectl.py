@@ -70,7 +70,8 @@ def load_all(kind, ns=None):
 def parse_flags(argv):
     """Split positionals from the flag subset we understand."""
     pos, flags = [], {"ns": "default", "all_ns": False, "label": None,
-                     "json": False, "patch": None, "subresource": None}
+                     "json": False, "patch": None, "subresource": None,
+                     "watch": False, "output": None}
     i = 0
     while i < len(argv):
         a = argv[i]
@@ -81,7 +82,10 @@ def parse_flags(argv):
         elif a == "-l" or a == "--selector":
             flags["label"] = argv[i + 1]; i += 2
         elif a == "-o":
-            flags["json"] = argv[i + 1] == "json"; i += 2
+            flags["json"] = argv[i + 1] == "json"
+            flags["output"] = argv[i + 1]; i += 2
+        elif a == "--watch" or a == "-w":
+            flags["watch"] = True; i += 1
         elif a == "-p":
             flags["patch"] = argv[i + 1]; i += 2
         elif a.startswith("--subresource"):
@@ -128,6 +132,19 @@ def cmd_get(argv):
             sys.stdout.write(f.read())
         return 0
     ns = None if flags["all_ns"] else flags["ns"]
+    if flags["watch"]:
+        # stream object names on every state change (the informer stand-in)
+        import time as _time
+
+        seen = {}
+        while True:
+            for o in load_all(kind, ns):
+                name = o.get("metadata", {}).get("name")
+                stamp = json.dumps(o, sort_keys=True)
+                if seen.get(name) != stamp:
+                    seen[name] = stamp
+                    print(f"{kind}/{name}", flush=True)
+            _time.sleep(0.1)
     items = load_all(kind, ns)
     if flags["label"]:
         k, v = flags["label"].split("=", 1)

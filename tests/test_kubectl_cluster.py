@@ -352,3 +352,29 @@ def test_shm_sizing_and_quantities():
     # 10Gi ≈ 10.7 G (decimal) -> floor to 10G
     assert dshm["emptyDir"]["sizeLimit"] == "10G"
     assert dshm["emptyDir"]["medium"] == "Memory"
+
+
+def test_watch_events_through_kubectl_stream(kube):
+    """The event-driven watcher consumes a real `kubectl get pods --watch`
+    subprocess stream end-to-end: a pod that becomes Ready after
+    subscription drains the watch without polling."""
+    import threading
+    import time
+
+    from dgl_operator_amd.operator_plane import watcher
+
+    c, state = kube
+    c.create_pod(Pod("w-0", "default",
+                     spec={"containers": [{"name": "c", "image": "i"}]},
+                     owner="j"))
+
+    def kubelet():
+        time.sleep(0.6)
+        set_pod_phase(state, "default", "w-0", "Running", ip="10.1.1.1")
+
+    t = threading.Thread(target=kubelet, daemon=True)
+    t.start()
+    t0 = time.time()
+    assert watcher.watch_events(c, "default", ["w-0"], "ready", timeout=10)
+    assert time.time() - t0 < 8
+    t.join()
