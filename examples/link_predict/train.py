@@ -59,7 +59,9 @@ def main():
     perm = torch.randperm(E, device=dev)
     n_test = E // 10
     test_e, train_e = perm[:n_test], perm[n_test:]
-    msg_g = Graph(src[train_e], dst[train_e], g.num_nodes)
+    # message graph = graph minus the held-out test edges, exactly the
+    # reference tutorial's dgl.remove_edges step (4_link_predict.py)
+    msg_g = g.remove_edges(test_e)
     gen = torch.Generator(device=dev).manual_seed(2)
 
     def neg_graph(n):
