@@ -37,10 +37,23 @@ def main():
     rg = g.reverse()
     print("reversed in_degrees:", rg.in_degrees().tolist())
 
-    # batching (graph classification input)
+    # batching (graph classification input) and its inverse
+    from dgl_operator_amd.graph import batch_num_edges, to_bidirected, unbatch
+
     graphs = [rmat_graph(6, 12, seed=i) for i in range(3)]
     bg, sizes = batch_graphs(graphs)
-    print(f"batched: {bg.num_nodes} nodes, batch_num_nodes={sizes.tolist()}")
+    print(f"batched: {bg.num_nodes} nodes, batch_num_nodes={sizes.tolist()}, "
+          f"batch_num_edges={batch_num_edges(bg, sizes).tolist()}")
+    parts = unbatch(bg, sizes)
+    print(f"unbatched back into {len(parts)} graphs of "
+          f"{[p.num_nodes for p in parts]} nodes")
+
+    # undirected view + isolated scratch space
+    bd = to_bidirected(g)
+    print(f"to_bidirected: {g.num_edges} -> {bd.num_edges} edges")
+    with g.local_scope():
+        g.ndata["scratch"] = torch.zeros(g.num_nodes)
+    print("scratch survives local_scope:", "scratch" in g.ndata)
 
     # device moves
     if torch.cuda.is_available():
