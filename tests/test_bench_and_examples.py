@@ -99,6 +99,7 @@ def test_bench_two_rank_driver_contract(halo):
      ["--epochs", "2", "--num-graphs", "20"]),
     ("examples/message_passing/train.py",
      ["--epochs", "5", "--nodes", "300", "--edges", "2000", "--feat", "16"]),
+    ("examples/graph_api/tour.py", []),
     ("examples/dgl_ke/train_ke.py",
      ["--max-step", "10", "--log-interval", "5", "--num-entities", "2000",
       "--num-relations", "10", "--num-triples", "5000", "--hidden-dim", "16",
