@@ -41,6 +41,18 @@ NID = "_ID"
 EID = "_ID"
 
 
+class _IntCompat(int):
+    """int that also accepts DGL's method-call spelling: DGL's
+    ``g.num_nodes()`` / ``g.num_edges()`` are METHODS, this package's are
+    properties — returning a callable int keeps both spellings working
+    for code migrated verbatim from the reference examples."""
+
+    __slots__ = ()
+
+    def __call__(self) -> int:
+        return int(self)
+
+
 class Graph:
     """Homogeneous directed graph. Edges point src -> dst; messages flow along edges."""
 
@@ -70,20 +82,20 @@ class Graph:
     # -- structure ---------------------------------------------------------
     @property
     def num_nodes(self) -> int:
-        return self._num_nodes
+        return _IntCompat(self._num_nodes)
 
     @property
     def num_edges(self) -> int:
-        return self._src.numel()
+        return _IntCompat(self._src.numel())
 
     # homogeneous graphs are their own src/dst sets (DGL API uniformity)
     @property
     def num_src_nodes(self) -> int:
-        return self._num_nodes
+        return _IntCompat(self._num_nodes)
 
     @property
     def num_dst_nodes(self) -> int:
-        return self._num_nodes
+        return _IntCompat(self._num_nodes)
 
     @property
     def device(self) -> torch.device:
@@ -307,15 +319,15 @@ class Block:
 
     @property
     def num_src_nodes(self) -> int:
-        return self._num_src
+        return _IntCompat(self._num_src)
 
     @property
     def num_dst_nodes(self) -> int:
-        return self._num_dst
+        return _IntCompat(self._num_dst)
 
     @property
     def num_edges(self) -> int:
-        return self.csc_indices.numel()
+        return _IntCompat(self.csc_indices.numel())
 
     @property
     def device(self) -> torch.device:

@@ -151,3 +151,19 @@ def test_dgl_api_conveniences():
     assert torch.equal(blk.srcdata[NID], torch.tensor([10, 20, 30]))
     assert torch.equal(blk.dstdata[NID], torch.tensor([10, 20]))
     assert doa.NID == "_ID"
+
+
+def test_num_nodes_both_spellings():
+    """DGL spells num_nodes/num_edges as METHODS; this package as
+    properties. Both must work on Graphs and Blocks (verbatim-migrated
+    reference code calls g.num_edges())."""
+    g = Graph(torch.tensor([0, 1]), torch.tensor([1, 2]), 3)
+    assert g.num_nodes == 3 and g.num_nodes() == 3
+    assert g.num_edges == 2 and g.num_edges() == 2
+    assert isinstance(g.num_nodes + 1, int)
+    from dgl_operator_amd.graph import Block
+
+    b = Block(torch.tensor([0, 1, 2]), torch.tensor([1, 2]), 3, 2,
+              srcdata_nids=torch.tensor([5, 6, 7]))
+    assert b.num_src_nodes == 3 and b.num_src_nodes() == 3
+    assert b.num_edges() == 2
