@@ -82,7 +82,11 @@ class DistGraph:
         ndata: Dict[str, torch.Tensor],  # owned rows only
     ):
         self.book = book
-        self.rank = rank
+        # _IntCompat: the reference calls g.rank() (train_dist.py uses the
+        # method spelling 7x); plain attribute reads keep working too
+        from ..graph.graph import _IntCompat
+
+        self.rank = _IntCompat(rank)
         self.lo, self.hi = book.owned_range(rank)
         self.csc_indptr = csc_indptr
         self.csc_indices = csc_indices
