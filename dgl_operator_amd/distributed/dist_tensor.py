@@ -25,6 +25,19 @@ from . import comm
 from .partition_book import PartitionBook
 
 
+def node_split(nodes: torch.Tensor, partition_book: PartitionBook,
+               rank: Optional[int] = None) -> torch.Tensor:
+    """Module-level dgl.distributed.node_split spelling (the reference
+    calls it with a GLOBAL boolean mask + the partition book,
+    train_dist.py:274-276): this rank's owned global ids where the mask
+    is set."""
+    if rank is None:
+        rank, _ = comm.world()
+    lo, hi = partition_book.owned_range(rank)
+    local = nodes[lo:hi].bool()
+    return torch.arange(lo, hi, device=nodes.device)[local]
+
+
 class DistTensor:
     """Row-sharded distributed tensor keyed by GLOBAL row id.
 
