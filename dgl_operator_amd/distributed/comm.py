@@ -27,6 +27,17 @@ def init_from_env(backend: Optional[str] = None) -> Tuple[int, int]:
     return dist.get_rank(), dist.get_world_size()
 
 
+def initialize(ip_config: Optional[str] = None, num_servers: int = 0,
+               num_workers: int = 0, backend: Optional[str] = None
+               ) -> Tuple[int, int]:
+    """dgl.distributed.initialize spelling (reference train_dist.py:267).
+    There are no graph-server or sampler processes to start here — the
+    equivalent setup is joining the RCCL/gloo process group; the ip_config
+    and process-count arguments are accepted and unused (torchrun's env
+    carries the rendezvous)."""
+    return init_from_env(backend=backend)
+
+
 def world() -> Tuple[int, int]:
     if dist.is_available() and dist.is_initialized():
         return dist.get_rank(), dist.get_world_size()
