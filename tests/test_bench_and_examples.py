@@ -36,6 +36,7 @@ def test_bench_json_contract():
     assert d["value"] > 0
     assert "edges/sec" in d["metric"]
     assert "global_batch" in d["config"] and "parallelism" in d["config"]
+    assert d["config"]["epoch_time_s"] > 0  # BASELINE companion number
 
 
 @pytest.mark.timeout(400)
