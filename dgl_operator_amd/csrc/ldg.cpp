@@ -20,6 +20,8 @@
 #include <ATen/Parallel.h>
 #include <torch/extension.h>
 
+#include <algorithm>
+#include <atomic>
 #include <cstdint>
 #include <vector>
 

@@ -25,7 +25,11 @@ import torch
 def _coo_to_compressed(
     rows: torch.Tensor, cols: torch.Tensor, num_rows: int
 ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
-    """Sort edges by ``rows`` and build indptr. Returns (indptr, sorted_cols, eids)."""
+    """Sort edges by ``rows`` and build indptr. Returns (indptr, sorted_cols,
+    eids). Measured round 2: torch's parallel radix argsort beats a native
+    counting-sort C++ builder ~2x at 100-400M edges on this 8-core host
+    (the row-bucketed scatter's full edge-list scans dominate), so the
+    torch path stays."""
     perm = torch.argsort(rows, stable=True)
     sorted_rows = rows[perm]
     indptr = torch.zeros(num_rows + 1, dtype=torch.int64, device=rows.device)
