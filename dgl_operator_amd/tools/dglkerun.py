@@ -55,13 +55,15 @@ def build_parser():
     p.add_argument("--batch-size", type=int, default=1024)
     p.add_argument("--neg-sample-size", type=int, default=256)
     p.add_argument("--max-step", type=int, default=1000)
-    p.add_argument("--regularization-coef", type=float, default=1e-9,
+    p.add_argument("--regularization-coef", "--regularization_coef",
+                   type=float, default=1e-9,
                    help="forwarded to the trainer "
                         "(reference dglkerun:301 fixed value)")
     p.add_argument("--test", action="store_true",
                    help="evaluate after training (reference dglkerun:300 "
                         "--test; maps to the trainer's --eval)")
-    p.add_argument("--batch-size-eval", type=int, default=1024)
+    p.add_argument("--batch-size-eval", "--batch_size_eval",
+                   type=int, default=1024)
     p.add_argument("--log-interval", type=int, default=100)
     p.add_argument("--save-path", default="ckpts")
     p.add_argument("--no-save-emb", action="store_true")

@@ -116,24 +116,25 @@ def _capture_ke_step(model, sampler, args, device):
 
 def main():
     p = argparse.ArgumentParser()
-    p.add_argument("--model-name", default="ComplEx")
-    p.add_argument("--hidden-dim", type=int, default=400)
+    p.add_argument("--model-name", "--model_name", default="ComplEx")
+    p.add_argument("--hidden-dim", "--hidden_dim", type=int, default=400)
     p.add_argument("--gamma", type=float, default=143.0)
     p.add_argument("--lr", type=float, default=0.1)
-    p.add_argument("--batch-size", type=int, default=1024)
-    p.add_argument("--neg-sample-size", type=int, default=256)
+    p.add_argument("--batch-size", "--batch_size", type=int, default=1024)
+    p.add_argument("--neg-sample-size", "--neg_sample_size", type=int,
+                   default=256)
     p.add_argument("--chunk-size", type=int, default=64)
     p.add_argument("--adversarial-temperature", "-adv-temp", type=float,
                    default=1.0, help="self-adversarial sampling temperature "
                                      "(dglke -adv is always on here)")
-    p.add_argument("--max-step", type=int, default=1000)
+    p.add_argument("--max-step", "--max_step", type=int, default=1000)
     p.add_argument("--regularization-coef", "--regularization_coef",
                    type=float, default=0.0,
                    help="Lp regularization of batch entity embeddings "
                         "(reference dglkerun passes 1e-9)")
     p.add_argument("--regularization-norm", "--regularization_norm",
                    type=int, default=3)
-    p.add_argument("--log-interval", type=int, default=100)
+    p.add_argument("--log-interval", "--log_interval", type=int, default=100)
     p.add_argument("--num-entities", type=int, default=1_000_000)
     p.add_argument("--num-relations", type=int, default=1000)
     p.add_argument("--num-triples", type=int, default=5_000_000)
@@ -143,8 +144,8 @@ def main():
                         "shard boundaries come from the partition json")
     p.add_argument("--dataset-name", default="synthetic",
                    help="dataset dir name under --data-path")
-    p.add_argument("--save-path", default="")
-    p.add_argument("--no-save-emb", action="store_true")
+    p.add_argument("--save-path", "--save_path", default="")
+    p.add_argument("--no-save-emb", "--no_save_emb", action="store_true")
     p.add_argument("--no-capture", dest="capture", action="store_false",
                    default=True,
                    help="disable hipGraph capture of the train step "
