@@ -348,3 +348,37 @@ def test_launch_cli_reference_contract(tmp_path, monkeypatch):
     for pod in ("job-worker-0", "job-worker-1"):
         assert (tmp_path / "pods" / pod / "w" /
                 "payload.txt").read_text() == "data"
+
+
+def test_dglrun_ignore_partition_reuse(tmp_path, monkeypatch):
+    """--ignore-partition reuses an existing partition (the reference
+    dglkerun PVC-reuse path applied to dglrun): the second partitioner
+    run must SKIP phase 1 and still deliver."""
+    import subprocess
+
+    REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    leadfile = tmp_path / "leadfile"
+    leadfile.write_text("127.0.0.1 30050 job-launcher slots=1\n")
+    env = dict(os.environ)
+    env["DGL_LOCAL_FABRIC_ROOT"] = str(tmp_path / "pods")
+    env["DGL_OPERATOR_PHASE_ENV"] = "Partitioner"
+    env["PYTHONPATH"] = REPO
+    argv = [sys.executable, "-m", "dgl_operator_amd.tools.dglrun",
+            "--graph-name", "toy", "--workspace", "ws",
+            "--leadfile", str(leadfile),
+            "--partition-entry-point",
+            os.path.join(REPO, "examples/graphsage_dist/load_and_partition_graph.py"),
+            "--partition-entry-args",
+            "--nodes 100 --edges 500 --feat 4 --classes 2 --algorithm range",
+            "--num-partitions", "2", "--ignore-partition"]
+    r1 = subprocess.run(argv, capture_output=True, text=True,
+                        cwd=str(tmp_path), env=env, timeout=120)
+    assert r1.returncode == 0, r1.stderr
+    assert "Phase 1/5 partition" in r1.stdout  # first run partitions
+    meta = tmp_path / "ws" / "dataset" / "toy.json"
+    mtime = meta.stat().st_mtime
+    r2 = subprocess.run(argv, capture_output=True, text=True,
+                        cwd=str(tmp_path), env=env, timeout=120)
+    assert r2.returncode == 0, r2.stderr
+    assert "Phase 1/5 skipped (--ignore-partition)" in r2.stdout
+    assert meta.stat().st_mtime == mtime  # untouched partition
