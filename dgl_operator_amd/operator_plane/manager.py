@@ -36,6 +36,8 @@ try:
         "dgljob_reconcile_duration_seconds", "Reconcile wall time"
     )
     JOB_PHASE = Gauge("dgljob_phase", "Job phase (1=active)", ["job", "phase"])
+    IS_LEADER = Gauge("dgljob_manager_is_leader",
+                      "1 while this replica holds the leader lease")
 except ImportError:  # pragma: no cover
     _PROM = False
 
@@ -99,6 +101,8 @@ class Manager:
         if self.is_leader != was:
             print(f"[manager] leadership {'acquired' if self.is_leader else 'lost'}"
                   f" ({self.identity})", flush=True)
+        if _PROM:
+            IS_LEADER.set(1.0 if self.is_leader else 0.0)
         return self.is_leader
 
     # -- job API (the CRD surface) -----------------------------------------
