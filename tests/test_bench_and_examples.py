@@ -144,6 +144,7 @@ def test_dglkerun_end_to_end(tmp_path):
              os.path.join(REPO, "examples/dgl_ke/train_ke.py"),
              "--model-name", "TransE_l2", "--hidden-dim", "16",
              "--batch-size", "64", "--neg-sample-size", "8", "--max-step", "20",
+             "--test", "--batch-size-eval", "32",
              "--save-path", "ckpts"],
             env={"DGL_LOCAL_FABRIC_ROOT": str(tmp_path / "pods"),
                  "PYTHONPATH": REPO},
