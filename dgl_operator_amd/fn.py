@@ -62,3 +62,10 @@ def mean(msg: str, out: str) -> ReduceFn:
 
 def u_dot_v(u: str, v: str, out: str) -> EdgeFn:
     return EdgeFn("u_dot_v", u, v, out)
+
+
+def u_add_v(u: str, v: str, out: str) -> EdgeFn:
+    """Per-edge src+dst field sum — the standard DGL spelling for GAT
+    attention logits (el[src] + er[dst]); the fused LeakyReLU variant is
+    ops.gat_score."""
+    return EdgeFn("u_add_v", u, v, out)

@@ -249,7 +249,15 @@ def _apply_edges_block(b, edge_func) -> None:
     from ..ops.udf import EdgeBatch
 
     if isinstance(edge_func, fnmod.EdgeFn):
-        assert edge_func.op == "u_dot_v"
+        if edge_func.op == "u_add_v":
+            src_l = b.csc_indices
+            dst_l = b.csc_dst()
+            b.edata[edge_func.out_field] = (
+                b.srcdata[edge_func.lhs_field][src_l]
+                + b.dstdata[edge_func.rhs_field][dst_l]
+            )
+            return
+        assert edge_func.op == "u_dot_v", edge_func.op
         b.edata[edge_func.out_field] = sddmm_dot(
             b, b.srcdata[edge_func.lhs_field], b.dstdata[edge_func.rhs_field]
         )
@@ -271,7 +279,14 @@ def _apply_edges(g, edge_func) -> None:
     from ..ops.udf import EdgeBatch
 
     if isinstance(edge_func, fnmod.EdgeFn):
-        assert edge_func.op == "u_dot_v"
+        if edge_func.op == "u_add_v":
+            s_, d_ = g.edges()
+            g.edata[edge_func.out_field] = (
+                g.ndata[edge_func.lhs_field][s_]
+                + g.ndata[edge_func.rhs_field][d_]
+            )
+            return
+        assert edge_func.op == "u_dot_v", edge_func.op
         g.edata[edge_func.out_field] = sddmm_dot(
             g, g.ndata[edge_func.lhs_field], g.ndata[edge_func.rhs_field]
         )

@@ -95,3 +95,27 @@ def test_block_fn_api():
     assert torch.allclose(blk.dstdata["h_N"], ref, atol=1e-6)
     blk.apply_edges(fn.u_dot_v("h", "h", "score"))
     assert blk.edata["score"].shape == (blk.num_edges,)
+
+
+def test_u_add_v_builtin():
+    """fn.u_add_v (GAT attention-logit spelling) on Graphs and Blocks
+    matches the explicit gather-add; Graph edata lands in original edge
+    order, Block edata in csc order like the other edge builtins."""
+    import dgl_operator_amd.fn as fn
+    from dgl_operator_amd.graph import Block, Graph
+
+    g = Graph(torch.tensor([0, 2, 1]), torch.tensor([1, 0, 2]), 3)
+    g.ndata["el"] = torch.tensor([1.0, 2.0, 3.0])
+    g.ndata["er"] = torch.tensor([10.0, 20.0, 30.0])
+    g.apply_edges(fn.u_add_v("el", "er", "e"))
+    s, d = g.edges()
+    assert torch.equal(g.edata["e"], g.ndata["el"][s] + g.ndata["er"][d])
+
+    blk = Block(torch.tensor([0, 1, 3]), torch.tensor([1, 0, 2]), 4, 2,
+                srcdata_nids=torch.tensor([7, 8, 9, 6]))
+    blk.srcdata["el"] = torch.tensor([1.0, 2.0, 3.0, 4.0])
+    blk.dstdata["er"] = torch.tensor([5.0, 6.0])
+    blk.apply_edges(fn.u_add_v("el", "er", "e"))
+    # csc order: dst0 edges [1], dst1 edges [0, 2]
+    expect = torch.tensor([2.0 + 5.0, 1.0 + 6.0, 3.0 + 6.0])
+    assert torch.equal(blk.edata["e"], expect)
