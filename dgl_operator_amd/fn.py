@@ -60,6 +60,10 @@ def mean(msg: str, out: str) -> ReduceFn:
     return ReduceFn("mean", msg, out)
 
 
+def max(msg: str, out: str) -> ReduceFn:  # noqa: A001 (DGL API name)
+    return ReduceFn("max", msg, out)
+
+
 def u_dot_v(u: str, v: str, out: str) -> EdgeFn:
     return EdgeFn("u_dot_v", u, v, out)
 

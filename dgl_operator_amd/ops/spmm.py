@@ -149,10 +149,30 @@ def gspmm(
     internally to CSC order.
     """
     assert op in ("copy_u", "u_mul_e")
-    assert reduce in ("sum", "mean")
+    assert reduce in ("sum", "mean", "max")
     w_csc = None
     if op == "u_mul_e":
         assert eweight is not None, "u_mul_e needs an edge weight"
         _, _, eids = gstruct.csc()
         w_csc = eweight[eids] if eids is not None else eweight
+    if reduce == "max":
+        # fn.max: eager scatter-amax (autograd routes gradients to the
+        # argmax source rows); zero-degree rows get 0 like DGL. Max is not
+        # a bandwidth-bound hot path in any reference workload, so no
+        # dedicated kernel.
+        indptr, indices, _ = gstruct.csc()
+        msgs = feat[indices]
+        if w_csc is not None:
+            w = w_csc.view(w_csc.shape + (1,) * (msgs.dim() - w_csc.dim()))
+            msgs = msgs * w
+        dst = _edge_dst(indptr)
+        num_dst = indptr.numel() - 1
+        init = torch.full((num_dst,) + msgs.shape[1:], float("-inf"),
+                          dtype=msgs.dtype, device=msgs.device)
+        idx = dst.view((-1,) + (1,) * (msgs.dim() - 1)).expand_as(msgs)
+        out = torch.scatter_reduce(init, 0, idx, msgs, reduce="amax",
+                                   include_self=True)
+        deg = (indptr[1:] - indptr[:-1]).view(
+            (-1,) + (1,) * (msgs.dim() - 1))
+        return torch.where(deg > 0, out, torch.zeros_like(out))
     return _GSpMM.apply(gstruct, feat, w_csc, reduce == "mean")

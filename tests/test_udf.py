@@ -119,3 +119,33 @@ def test_u_add_v_builtin():
     # csc order: dst0 edges [1], dst1 edges [0, 2]
     expect = torch.tensor([2.0 + 5.0, 1.0 + 6.0, 3.0 + 6.0])
     assert torch.equal(blk.edata["e"], expect)
+
+
+def test_fn_max_reducer():
+    """fn.max builtin: per-dst max over in-edge messages, zero rows for
+    zero-degree dsts (DGL semantics), gradients flow to argmax sources."""
+    import dgl_operator_amd.fn as fn
+    from dgl_operator_amd.graph import Graph
+
+    g = Graph(torch.tensor([0, 1, 2, 0]), torch.tensor([1, 2, 1, 2]), 4)
+    x = torch.tensor([[1.0, 9.0], [5.0, 2.0], [3.0, 3.0], [0.0, 0.0]],
+                     requires_grad=True)
+    g.ndata["h"] = x
+    g.update_all(fn.copy_u("h", "m"), fn.max("m", "out"))
+    out = g.ndata["out"]
+    # dst1 <- {src0, src2}: max = [3, 9]; dst2 <- {src1, src0}: [5, 9]
+    assert torch.equal(out.detach(),
+                       torch.tensor([[0.0, 0.0], [3.0, 9.0], [5.0, 9.0],
+                                     [0.0, 0.0]]))
+    out.sum().backward()
+    # argmax sources get gradient 1: src0 col1 twice, src2 col0, src1 col0
+    assert torch.equal(x.grad,
+                       torch.tensor([[0.0, 2.0], [1.0, 0.0], [1.0, 0.0],
+                                     [0.0, 0.0]]))
+    # weighted variant
+    g2 = Graph(torch.tensor([0, 1]), torch.tensor([2, 2]), 3)
+    g2.ndata["h"] = torch.tensor([[2.0], [4.0]])
+    g2.edata["w"] = torch.tensor([10.0, 1.0])
+    g2.update_all(fn.u_mul_e("h", "w", "m"), fn.max("m", "out"))
+    assert torch.equal(g2.ndata["out"],
+                       torch.tensor([[0.0], [0.0], [20.0]]))
