@@ -73,3 +73,18 @@ def u_add_v(u: str, v: str, out: str) -> EdgeFn:
     attention logits (el[src] + er[dst]); the fused LeakyReLU variant is
     ops.gat_score."""
     return EdgeFn("u_add_v", u, v, out)
+
+
+__all__ = [
+    "MessageFn",
+    "ReduceFn",
+    "EdgeFn",
+    "copy_u",
+    "copy_src",
+    "u_mul_e",
+    "sum",
+    "mean",
+    "max",
+    "u_dot_v",
+    "u_add_v",
+]
